@@ -1,0 +1,153 @@
+"""Checkpoint transport tests (reference: torchft/checkpointing/*_test.py)."""
+
+import io
+import threading
+import time
+from datetime import timedelta
+
+import pytest
+import torch
+
+from torchft_amd.checkpointing import HTTPTransport, RWLock
+from torchft_amd.checkpointing._serialization import streaming_load, streaming_save
+
+
+class TestSerialization:
+    def test_roundtrip_nested(self):
+        obj = {
+            "model": {
+                "w": torch.randn(4, 8),
+                "b": torch.arange(10, dtype=torch.int64),
+                "half": torch.randn(3, 3).to(torch.bfloat16),
+            },
+            "step": 7,
+            "names": ["a", "b"],
+            "empty": torch.empty(0),
+            "noncontig": torch.randn(6, 6).t(),
+        }
+        buf = io.BytesIO()
+        streaming_save(obj, buf)
+        buf.seek(0)
+        loaded = streaming_load(buf)
+        assert loaded["step"] == 7
+        assert loaded["names"] == ["a", "b"]
+        torch.testing.assert_close(loaded["model"]["w"], obj["model"]["w"])
+        torch.testing.assert_close(loaded["model"]["b"], obj["model"]["b"])
+        torch.testing.assert_close(loaded["model"]["half"], obj["model"]["half"])
+        torch.testing.assert_close(loaded["noncontig"], obj["noncontig"].contiguous())
+        assert loaded["empty"].numel() == 0
+
+    def test_scalar_and_none_leaves(self):
+        obj = {"lr": 0.1, "none": None, "t": torch.ones(2)}
+        buf = io.BytesIO()
+        streaming_save(obj, buf)
+        buf.seek(0)
+        loaded = streaming_load(buf)
+        assert loaded["lr"] == 0.1
+        assert loaded["none"] is None
+
+
+class TestRWLock:
+    def test_concurrent_readers(self):
+        lock = RWLock(timeout=5)
+        with lock.r_lock():
+            with lock.r_lock():
+                pass
+
+    def test_writer_excludes_readers(self):
+        lock = RWLock(timeout=5)
+        lock.w_acquire()
+        with pytest.raises(TimeoutError):
+            lock.r_acquire(timeout=0.1)
+        lock.w_release()
+        with lock.r_lock():
+            pass
+
+    def test_writer_waits_for_readers(self):
+        lock = RWLock(timeout=5)
+        lock.r_acquire()
+        acquired = threading.Event()
+
+        def writer():
+            lock.w_acquire()
+            acquired.set()
+            lock.w_release()
+
+        t = threading.Thread(target=writer)
+        t.start()
+        time.sleep(0.1)
+        assert not acquired.is_set()
+        lock.r_release()
+        t.join(timeout=5)
+        assert acquired.is_set()
+
+
+class TestHTTPTransport:
+    def test_send_recv(self):
+        src = HTTPTransport(timeout=timedelta(seconds=10))
+        dst = HTTPTransport(timeout=timedelta(seconds=10))
+        try:
+            sd = {"user": {"default": {"w": torch.randn(16, 16)}}, "torchft": {"step": 3}}
+            src.send_checkpoint([1], step=3, state_dict=sd, timeout=timedelta(seconds=10))
+            got = dst.recv_checkpoint(
+                src_rank=0, metadata=src.metadata(), step=3, timeout=timedelta(seconds=10)
+            )
+            torch.testing.assert_close(got["user"]["default"]["w"], sd["user"]["default"]["w"])
+            assert got["torchft"]["step"] == 3
+        finally:
+            src.shutdown()
+            dst.shutdown()
+
+    def test_wrong_step_rejected(self):
+        src = HTTPTransport(timeout=timedelta(seconds=10))
+        dst = HTTPTransport(timeout=timedelta(seconds=10))
+        try:
+            src.send_checkpoint(
+                [1], step=3, state_dict={"x": 1}, timeout=timedelta(seconds=10)
+            )
+            with pytest.raises(Exception):
+                dst.recv_checkpoint(
+                    src_rank=0,
+                    metadata=src.metadata(),
+                    step=99,
+                    timeout=timedelta(seconds=5),
+                )
+        finally:
+            src.shutdown()
+            dst.shutdown()
+
+    def test_disallow_blocks_serving(self):
+        src = HTTPTransport(timeout=timedelta(seconds=10))
+        dst = HTTPTransport(timeout=timedelta(seconds=10))
+        try:
+            src.send_checkpoint(
+                [1], step=3, state_dict={"x": 1}, timeout=timedelta(seconds=10)
+            )
+            src.disallow_checkpoint()
+            with pytest.raises(Exception):
+                dst.recv_checkpoint(
+                    src_rank=0,
+                    metadata=src.metadata(),
+                    step=3,
+                    timeout=timedelta(seconds=5),
+                )
+        finally:
+            src.shutdown()
+            dst.shutdown()
+
+    def test_chunked_fetch(self):
+        src = HTTPTransport(timeout=timedelta(seconds=10), num_chunks=3)
+        dst = HTTPTransport(timeout=timedelta(seconds=10), num_chunks=3)
+        try:
+            sd = {f"t{i}": torch.randn(32) for i in range(7)}
+            sd["meta"] = "hello"
+            src.send_checkpoint([1], step=1, state_dict=sd, timeout=timedelta(seconds=10))
+            got = dst.recv_checkpoint(
+                src_rank=0, metadata=src.metadata(), step=1, timeout=timedelta(seconds=10)
+            )
+            for i in range(7):
+                torch.testing.assert_close(got[f"t{i}"], sd[f"t{i}"])
+            assert got["meta"] == "hello"
+        finally:
+            src.shutdown()
+            dst.shutdown()

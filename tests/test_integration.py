@@ -1,0 +1,160 @@
+"""Multi-replica fault-tolerance integration tests.
+
+Reference strategy: torchft/manager_integ_test.py — a real LighthouseServer
+plus N replica groups (threads), injected failures, replicas restart, and at
+the end all replica state dicts must agree bitwise.
+"""
+
+import logging
+import threading
+from concurrent.futures import ThreadPoolExecutor
+from dataclasses import dataclass, field
+from datetime import timedelta
+from typing import Dict, List, Optional
+
+import pytest
+import torch
+import torch.nn as nn
+from torch.distributed import TCPStore
+
+from torchft_amd._ftcore import LighthouseServer
+from torchft_amd.ddp import DistributedDataParallel
+from torchft_amd.manager import Manager
+from torchft_amd.optim import OptimizerWrapper
+from torchft_amd.process_group import FakeProcessGroupWrapper, ProcessGroupGloo
+
+logging.basicConfig(level=logging.WARNING)
+
+
+class InjectedFailure(Exception):
+    pass
+
+
+@dataclass
+class EventInjector:
+    """Inject failures at (replica, step): raise before the step, or error
+    the next allreduce future (reference: manager_integ_test.py:99-177)."""
+
+    failures: Dict[int, int] = field(default_factory=dict)  # replica -> step
+    allreduce_failures: Dict[int, int] = field(default_factory=dict)
+    count: int = 0
+
+    def fail_at(self, replica: int, step: int) -> "EventInjector":
+        self.failures[replica] = step
+        return self
+
+    def fail_allreduce_at(self, replica: int, step: int) -> "EventInjector":
+        self.allreduce_failures[replica] = step
+        return self
+
+    def check(self, replica: int, step: int, pg: FakeProcessGroupWrapper) -> None:
+        if self.failures.get(replica) == step:
+            del self.failures[replica]
+            self.count += 1
+            raise InjectedFailure(f"injected failure at replica {replica} step {step}")
+        if self.allreduce_failures.get(replica) == step:
+            del self.allreduce_failures[replica]
+            self.count += 1
+            pg.report_future_error(RuntimeError("injected allreduce failure"))
+
+
+def _make_model() -> nn.Module:
+    torch.manual_seed(42)
+    return nn.Sequential(nn.Linear(8, 16), nn.ReLU(), nn.Linear(16, 4))
+
+
+def _replica_main(
+    replica_id: int,
+    lighthouse_addr: str,
+    injector: EventInjector,
+    total_steps: int,
+    attempts: int = 3,
+) -> Dict[str, torch.Tensor]:
+    """Run one replica group (world_size=1) to total_steps, restarting on
+    injected failures, and return the final model state dict."""
+    for attempt in range(attempts):
+        store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+        model = _make_model()
+        pg = FakeProcessGroupWrapper(ProcessGroupGloo(timeout=timedelta(seconds=20)))
+        manager = Manager(
+            pg=pg,
+            load_state_dict=model.load_state_dict,
+            state_dict=model.state_dict,
+            min_replica_size=1,
+            rank=0,
+            world_size=1,
+            store_addr="127.0.0.1",
+            store_port=store.port,
+            lighthouse_addr=lighthouse_addr,
+            replica_id=f"replica_{replica_id}",
+            hostname="127.0.0.1",
+            timeout=timedelta(seconds=20),
+            quorum_timeout=timedelta(seconds=20),
+            connect_timeout=timedelta(seconds=10),
+        )
+        try:
+            ddp = DistributedDataParallel(manager, model)
+            opt = OptimizerWrapper(manager, torch.optim.SGD(model.parameters(), lr=0.05))
+            criterion = nn.MSELoss()
+            while manager.current_step() < total_steps:
+                injector.check(replica_id, manager.current_step(), pg)
+                torch.manual_seed(manager.current_step())  # same data all replicas
+                x = torch.randn(4, 8)
+                y = torch.randn(4, 4)
+                opt.zero_grad()
+                loss = criterion(ddp(x), y)
+                loss.backward()
+                opt.step()
+            return {k: v.detach().clone() for k, v in model.state_dict().items()}
+        except InjectedFailure:
+            if attempt == attempts - 1:
+                raise
+            continue
+        finally:
+            manager.shutdown(wait=False)
+    raise RuntimeError("unreachable")
+
+
+def _run_replicas(
+    num_replicas: int, total_steps: int, injector: EventInjector, min_replicas: int = 1
+) -> List[Dict[str, torch.Tensor]]:
+    lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=min_replicas, join_timeout_ms=100)
+    try:
+        with ThreadPoolExecutor(max_workers=num_replicas) as ex:
+            futs = [
+                ex.submit(_replica_main, i, lh.address(), injector, total_steps)
+                for i in range(num_replicas)
+            ]
+            return [f.result(timeout=120) for f in futs]
+    finally:
+        lh.shutdown()
+
+
+def assert_state_dicts_equal(dicts: List[Dict[str, torch.Tensor]]) -> None:
+    for other in dicts[1:]:
+        for k, v in dicts[0].items():
+            torch.testing.assert_close(v, other[k], rtol=0, atol=0, msg=f"mismatch at {k}")
+
+
+class TestFTIntegration:
+    def test_healthy_two_replicas(self):
+        dicts = _run_replicas(2, total_steps=5, injector=EventInjector())
+        assert_state_dicts_equal(dicts)
+
+    def test_replica_failure_and_recovery(self):
+        injector = EventInjector().fail_at(replica=1, step=2)
+        dicts = _run_replicas(2, total_steps=6, injector=injector)
+        assert injector.count == 1
+        assert_state_dicts_equal(dicts)
+
+    def test_allreduce_failure_recovers(self):
+        injector = EventInjector().fail_allreduce_at(replica=0, step=2)
+        dicts = _run_replicas(2, total_steps=6, injector=injector)
+        assert injector.count == 1
+        assert_state_dicts_equal(dicts)
+
+    def test_three_replicas_two_failures(self):
+        injector = EventInjector().fail_at(replica=0, step=2).fail_at(replica=2, step=3)
+        dicts = _run_replicas(3, total_steps=7, injector=injector)
+        assert injector.count == 2
+        assert_state_dicts_equal(dicts)

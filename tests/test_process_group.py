@@ -1,0 +1,181 @@
+"""Process-group layer tests (reference strategy: torchft/process_group_test.py
+— per-backend collective table, threads-as-ranks with a port-0 TCPStore)."""
+
+import threading
+from concurrent.futures import ThreadPoolExecutor
+from datetime import timedelta
+
+import pytest
+import torch
+from torch.distributed import ReduceOp, TCPStore
+from torch.distributed.distributed_c10d import (
+    AllgatherOptions,
+    AllreduceOptions,
+    AllToAllOptions,
+    BroadcastOptions,
+    ReduceScatterOptions,
+)
+
+from torchft_amd.process_group import (
+    ErrorSwallowingProcessGroupWrapper,
+    FakeProcessGroupWrapper,
+    ProcessGroupDummy,
+    ProcessGroupGloo,
+    _DummyWork,
+)
+
+
+class TestProcessGroupDummy:
+    def test_allreduce_passthrough(self):
+        pg = ProcessGroupDummy(0, 1)
+        t = torch.ones(4)
+        work = pg.allreduce([t], AllreduceOptions())
+        assert work.wait()
+        torch.testing.assert_close(t, torch.ones(4))
+
+    def test_allgather_copies(self):
+        pg = ProcessGroupDummy(0, 1)
+        inp = [torch.arange(4.0)]
+        out = [[torch.zeros(4)]]
+        pg.allgather(out, inp, AllgatherOptions()).wait()
+        torch.testing.assert_close(out[0][0], inp[0])
+
+    def test_alltoall_copies(self):
+        pg = ProcessGroupDummy(0, 1)
+        inp = torch.arange(6.0)
+        out = torch.zeros(6)
+        pg.alltoall_base(out, inp, [], [], AllToAllOptions()).wait()
+        torch.testing.assert_close(out, inp)
+
+    def test_reduce_scatter_copies(self):
+        pg = ProcessGroupDummy(0, 1)
+        out = [torch.zeros(3)]
+        inp = [[torch.arange(3.0)]]
+        pg.reduce_scatter(out, inp, ReduceScatterOptions()).wait()
+        torch.testing.assert_close(out[0], inp[0][0])
+
+    def test_configure_counts(self):
+        pg = ProcessGroupDummy(0, 1)
+        pg.configure("addr", "0", 0, 1)
+        assert pg.configure_count == 1
+
+    def test_size_and_backend(self):
+        pg = ProcessGroupDummy(0, 1)
+        assert pg.size() == 1
+        assert pg.getBackendName() == "torchft-dummy"
+
+
+def _run_ranks(world_size, fn):
+    """Threads-as-ranks helper with a shared port-0 TCPStore."""
+    store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+    store_addr = f"127.0.0.1:{store.port}/test"
+    with ThreadPoolExecutor(max_workers=world_size) as ex:
+        futs = [ex.submit(fn, store_addr, rank, world_size) for rank in range(world_size)]
+        return [f.result(timeout=60) for f in futs]
+
+
+class TestProcessGroupGloo:
+    def test_allreduce_two_ranks(self):
+        def run(store_addr, rank, world):
+            pg = ProcessGroupGloo(timeout=timedelta(seconds=20))
+            pg.configure(store_addr, f"r{rank}", rank, world)
+            t = torch.full((8,), float(rank + 1))
+            opts = AllreduceOptions()
+            opts.reduceOp = ReduceOp.SUM
+            pg.allreduce([t], opts).wait()
+            return t
+
+        results = _run_ranks(2, run)
+        for t in results:
+            torch.testing.assert_close(t, torch.full((8,), 3.0))
+
+    def test_broadcast_and_allgather(self):
+        def run(store_addr, rank, world):
+            pg = ProcessGroupGloo(timeout=timedelta(seconds=20))
+            pg.configure(store_addr, f"r{rank}", rank, world)
+            t = torch.full((4,), float(rank))
+            opts = BroadcastOptions()
+            opts.rootRank = 1
+            pg.broadcast([t], opts).wait()
+
+            out = [[torch.zeros(4) for _ in range(world)]]
+            pg.allgather(out, [t], AllgatherOptions()).wait()
+            return t, out
+
+        for t, out in _run_ranks(2, run):
+            torch.testing.assert_close(t, torch.full((4,), 1.0))
+            for o in out[0]:
+                torch.testing.assert_close(o, torch.full((4,), 1.0))
+
+    def test_send_recv(self):
+        def run(store_addr, rank, world):
+            pg = ProcessGroupGloo(timeout=timedelta(seconds=20))
+            pg.configure(store_addr, f"r{rank}", rank, world)
+            if rank == 0:
+                t = torch.arange(5.0)
+                pg.send([t], 1, tag=7).wait()
+                return t
+            t = torch.zeros(5)
+            pg.recv([t], 0, tag=7).wait()
+            return t
+
+        r = _run_ranks(2, run)
+        torch.testing.assert_close(r[0], r[1])
+
+    def test_reconfigure_shrinks_world(self):
+        # reconfiguring with a new prefix builds a fresh communicator
+        store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+
+        def run(rank, world):
+            pg = ProcessGroupGloo(timeout=timedelta(seconds=20))
+            pg.configure(f"127.0.0.1:{store.port}/q0", f"r{rank}", rank, world)
+            t = torch.ones(2)
+            opts = AllreduceOptions()
+            opts.reduceOp = ReduceOp.SUM
+            pg.allreduce([t], opts).wait()
+            assert t[0].item() == world
+            return pg
+
+        with ThreadPoolExecutor(max_workers=2) as ex:
+            pgs = list(ex.map(lambda r: run(r, 2), range(2)))
+
+        # shrink to world of 1 (only rank 0 survives)
+        pg0 = pgs[0]
+        pg0.configure(f"127.0.0.1:{store.port}/q1", "r0", 0, 1)
+        t = torch.ones(2)
+        opts = AllreduceOptions()
+        opts.reduceOp = ReduceOp.SUM
+        pg0.allreduce([t], opts).wait()
+        assert t[0].item() == 1
+
+    def test_reduce_scatter_rejected(self):
+        pg = ProcessGroupGloo()
+        with pytest.raises(RuntimeError, match="does not support"):
+            pg.reduce_scatter([], [], ReduceScatterOptions())
+
+
+class TestWrappers:
+    def test_error_swallowing(self):
+        inner = ProcessGroupDummy(0, 1)
+        pg = ErrorSwallowingProcessGroupWrapper(inner)
+        t = torch.ones(2)
+        work = pg.allreduce([t], AllreduceOptions())
+        assert work.wait()
+        assert pg.error() is None
+
+        pg.report_error(RuntimeError("boom"))
+        work = pg.allreduce([t], AllreduceOptions())
+        assert isinstance(work, _DummyWork)
+        # configure clears the error
+        pg.configure("addr", "0", 0, 1)
+        assert pg.error() is None
+
+    def test_fake_future_error(self):
+        inner = ProcessGroupDummy(0, 1)
+        pg = FakeProcessGroupWrapper(inner)
+        pg.report_future_error(RuntimeError("injected"))
+        t = torch.ones(2)
+        work = pg.allreduce([t], AllreduceOptions())
+        fut = work.get_future()
+        with pytest.raises(RuntimeError, match="injected"):
+            fut.wait()

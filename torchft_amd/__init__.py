@@ -1,9 +1,42 @@
 """torchft_amd: MI355X-native per-step fault-tolerant training framework.
 
-Capability parity with meta-pytorch/torchft (reference layer map in
-SURVEY.md), rebuilt for AMD Instinct MI355X: C++ coordination services,
+Capability parity with meta-pytorch/torchft (layer map in SURVEY.md), rebuilt
+for AMD Instinct MI355X: C++ coordination services (lighthouse + manager),
 RCCL-over-xGMI reconfigurable process groups, CDNA4 HIP kernels for the
-quantized collectives and fused optimizer steps.
+fp8-quantized collectives and fused optimizer steps, and HIP-stream-based
+live healing.
 """
 
+from torchft_amd.data import DistributedSampler
+from torchft_amd.ddp import DistributedDataParallel
+from torchft_amd.manager import Manager, WorldSizeMode
+from torchft_amd.optim import OptimizerWrapper
+from torchft_amd.process_group import (
+    ProcessGroup,
+    ProcessGroupDummy,
+    ProcessGroupGloo,
+    ProcessGroupNCCL,
+    ProcessGroupRCCL,
+)
+
 __version__ = "0.1.0"
+
+__all__ = [
+    "DistributedDataParallel",
+    "DistributedSampler",
+    "Manager",
+    "OptimizerWrapper",
+    "ProcessGroup",
+    "ProcessGroupDummy",
+    "ProcessGroupGloo",
+    "ProcessGroupNCCL",
+    "ProcessGroupRCCL",
+    "WorldSizeMode",
+]
+
+try:  # LocalSGD/DiLoCo are available once local_sgd is importable
+    from torchft_amd.local_sgd import DiLoCo, LocalSGD  # noqa: F401
+
+    __all__ += ["DiLoCo", "LocalSGD"]
+except ImportError:  # pragma: no cover - during incremental builds
+    pass

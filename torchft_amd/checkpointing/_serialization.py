@@ -1,0 +1,156 @@
+"""Streaming state-dict serialization for checkpoint transports.
+
+Non-seekable stream format (sockets/HTTP bodies): a pickled header with the
+pytree spec + per-leaf metadata, followed by each tensor's raw bytes in
+order. Tensors never round-trip through torch.save's zip container, so a 100
+GB-class MI355X state dict streams at wire speed with one staging copy.
+
+Reference parity: torchft/checkpointing/_serialization.py (which delegates to
+torch.distributed._serialization._streaming_save/load) and the
+_TensorMeta/_DTensorMeta scheme of torchft/checkpointing/pg_transport.py.
+"""
+
+from __future__ import annotations
+
+import pickle
+import struct
+from dataclasses import dataclass
+from typing import IO, Any, List, Tuple, Union
+
+import torch
+from torch.utils._pytree import TreeSpec, tree_flatten, tree_unflatten
+
+try:
+    from torch.distributed.tensor import DTensor
+
+    HAS_DTENSOR = True
+except ImportError:  # pragma: no cover
+    HAS_DTENSOR = False
+
+
+@dataclass
+class _TensorMeta:
+    shape: torch.Size
+    dtype: torch.dtype
+    nbytes: int
+
+
+@dataclass
+class _DTensorMeta:
+    local: _TensorMeta
+    spec_bytes: bytes  # pickled DTensorSpec
+
+
+@dataclass
+class _PickledLeaf:
+    data: bytes
+
+
+LeafMeta = Union[_TensorMeta, _DTensorMeta, _PickledLeaf]
+
+
+def _tensor_bytes(t: torch.Tensor) -> torch.Tensor:
+    """Flat uint8 CPU view of a tensor's data (contiguous copy if needed)."""
+    t = t.detach()
+    if not t.is_contiguous():
+        t = t.contiguous()
+    if t.device.type != "cpu":
+        t = t.cpu()
+    if t.numel() == 0:
+        return torch.empty(0, dtype=torch.uint8)
+    return t.view(-1).view(torch.uint8)
+
+
+def _meta_for(t: torch.Tensor) -> _TensorMeta:
+    return _TensorMeta(
+        shape=t.shape, dtype=t.dtype, nbytes=t.numel() * t.element_size()
+    )
+
+
+def split_state_dict(obj: Any) -> Tuple[List[LeafMeta], List[torch.Tensor], TreeSpec]:
+    """Flatten ``obj``; return (metas, tensors-to-stream, treespec).
+
+    ``tensors`` holds only the tensor leaves (local tensors for DTensor), in
+    meta order for the tensor-typed metas.
+    """
+    leaves, spec = tree_flatten(obj)
+    metas: List[LeafMeta] = []
+    tensors: List[torch.Tensor] = []
+    for leaf in leaves:
+        if HAS_DTENSOR and isinstance(leaf, DTensor):
+            local = leaf._local_tensor
+            metas.append(
+                _DTensorMeta(local=_meta_for(local), spec_bytes=pickle.dumps(leaf._spec))
+            )
+            tensors.append(local)
+        elif isinstance(leaf, torch.Tensor):
+            metas.append(_meta_for(leaf))
+            tensors.append(leaf)
+        else:
+            metas.append(_PickledLeaf(data=pickle.dumps(leaf)))
+    return metas, tensors, spec
+
+
+def streaming_save(obj: Any, f: IO[bytes]) -> None:
+    metas, tensors, spec = split_state_dict(obj)
+    header = pickle.dumps((spec, metas))
+    f.write(struct.pack("<Q", len(header)))
+    f.write(header)
+    for t in tensors:
+        raw = _tensor_bytes(t)
+        f.write(raw.numpy().tobytes() if raw.numel() else b"")
+
+
+def _read_exact(f: IO[bytes], n: int) -> bytes:
+    chunks = []
+    got = 0
+    while got < n:
+        chunk = f.read(n - got)
+        if not chunk:
+            raise EOFError(f"stream ended after {got}/{n} bytes")
+        chunks.append(chunk)
+        got += len(chunk)
+    return b"".join(chunks)
+
+
+def _readinto_tensor(f: IO[bytes], meta: _TensorMeta) -> torch.Tensor:
+    t = torch.empty(meta.shape, dtype=meta.dtype)
+    if meta.nbytes:
+        buf = t.view(-1).view(torch.uint8).numpy()
+        view = memoryview(buf)
+        got = 0
+        while got < meta.nbytes:
+            k = f.readinto(view[got:]) if hasattr(f, "readinto") else None
+            if k is None:
+                data = f.read(meta.nbytes - got)
+                if not data:
+                    raise EOFError("stream ended mid-tensor")
+                view[got : got + len(data)] = data
+                k = len(data)
+            elif k == 0:
+                raise EOFError("stream ended mid-tensor")
+            got += k
+    return t
+
+
+def streaming_load(f: IO[bytes]) -> Any:
+    (header_len,) = struct.unpack("<Q", _read_exact(f, 8))
+    spec, metas = pickle.loads(_read_exact(f, header_len))
+    leaves: List[Any] = []
+    for meta in metas:
+        if isinstance(meta, _PickledLeaf):
+            leaves.append(pickle.loads(meta.data))
+        elif isinstance(meta, _DTensorMeta):
+            local = _readinto_tensor(f, meta.local)
+            dt_spec = pickle.loads(meta.spec_bytes)
+            assert HAS_DTENSOR
+            leaves.append(
+                DTensor(
+                    local.requires_grad_(False),
+                    dt_spec,
+                    requires_grad=False,
+                )
+            )
+        else:
+            leaves.append(_readinto_tensor(f, meta))
+    return tree_unflatten(leaves, spec)

@@ -1,0 +1,316 @@
+"""Flagship benchmark: fault-tolerant Llama-3 training goodput on MI355X.
+
+Measures the BASELINE.json metric — goodput tokens/sec for Llama-3-8B under
+the full fault-tolerance stack (per-step quorum + should_commit through the
+C++ lighthouse/manager services, gradient allreduce across replica groups
+over RCCL) — on synthetic token data with random-init weights.
+
+Topology on N GPUs of one node (weak scaling, fixed per-GPU batch):
+  N=1 -> 1 replica x 1 shard        N=2 -> 2 replicas x 1 shard (FT-DDP)
+  N=4 -> 2 replicas x 2 shards      N=8 -> 2 replicas x 4 shards (FT-HSDP)
+Cross-replica gradient reduction goes through Manager.allreduce (the FT
+path); intra-group sharding is FSDP2 fully_shard over the group's RCCL mesh
+with set_all_reduce_hook routing the cross-replica allreduce to the Manager.
+
+Launch (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+from datetime import timedelta
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def parse_args() -> argparse.Namespace:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--model", default="llama3_8b",
+                   choices=["llama3_8b", "llama3_70b", "debug"])
+    p.add_argument("--batch", type=int, default=2, help="per-GPU batch size")
+    p.add_argument("--seq", type=int, default=8192)
+    p.add_argument("--no-ft", action="store_true",
+                   help="disable the fault-tolerance control plane (raw perf)")
+    p.add_argument("--quantize", action="store_true",
+                   help="fp8-quantized cross-replica allreduce")
+    return p.parse_args()
+
+
+def main() -> None:
+    args = parse_args()
+
+    def dbg(msg: str) -> None:
+        if os.environ.get("TFT_BENCH_DEBUG"):
+            print(f"[bench dbg rank={os.environ.get('RANK','0')}] {msg}",
+                  file=sys.stderr, flush=True)
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    master_addr = os.environ.get("MASTER_ADDR", "127.0.0.1")
+    master_port = int(os.environ.get("MASTER_PORT", "29500"))
+    assert world == args.gpus or world == 1, (
+        f"WORLD_SIZE={world} does not match --gpus={args.gpus}"
+    )
+    world = max(world, 1)
+
+    use_cuda = torch.cuda.is_available()
+    device = torch.device("cuda", local_rank) if use_cuda else torch.device("cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+
+    # topology: up to 2 replica groups, shards within a group
+    replicas = min(2, world) if world > 1 else 1
+    shards = world // replicas
+    group = rank // shards
+    group_rank = rank % shards
+
+    from torch.distributed import TCPStore
+
+    # bootstrap store for global coordination (lighthouse addr, barrier PG)
+    boot_port = master_port + 1
+    boot_store = TCPStore(
+        master_addr, boot_port, world_size=world, is_master=(rank == 0),
+        wait_for_workers=False, timeout=timedelta(seconds=120),
+    )
+
+    from torchft_amd.process_group import ProcessGroupGloo
+
+    # global gloo PG for timing barriers only (not part of the FT path)
+    barrier_pg = ProcessGroupGloo(timeout=timedelta(seconds=120))
+    barrier_pg.configure(f"{master_addr}:{boot_port}/bench_barrier", str(rank), rank, world)
+
+    dbg("barrier pg configured")
+
+    def global_barrier() -> None:
+        from torch.distributed.distributed_c10d import BarrierOptions
+
+        barrier_pg.barrier(BarrierOptions()).wait()
+
+    # per-group default process group (for the FSDP mesh; only when sharding).
+    # Explicit tcp:// init — torchrun's TORCHELASTIC_USE_AGENT_STORE would
+    # otherwise point every group at the same agent store.
+    group_port = master_port + 10 + group
+    import torch.distributed as dist
+
+    dbg(f"init group pg shards={shards} group={group}")
+    if shards > 1:
+        # explicit store: under torchrun, TORCHELASTIC_USE_AGENT_STORE makes
+        # tcp:// rendezvous assume the store already exists and hang
+        group_store = TCPStore(
+            master_addr, group_port, is_master=(group_rank == 0),
+            wait_for_workers=False, timeout=timedelta(seconds=120),
+        )
+        dist.init_process_group(
+            "nccl" if use_cuda else "gloo",
+            store=dist.PrefixStore("group_pg", group_store),
+            rank=group_rank,
+            world_size=shards,
+        )
+
+    dbg("group pg init done")
+    # dedicated TCPStore per replica group for the Manager (leader hosts it)
+    mgr_port = master_port + 100 + group
+    _mgr_store_server = (
+        TCPStore(master_addr, mgr_port, is_master=True, wait_for_workers=False)
+        if group_rank == 0
+        else None
+    )
+
+    # ---- model ------------------------------------------------------------
+    from torchft_amd.models import LLAMA3_8B, LLAMA3_70B, LLAMA_DEBUG, Llama
+
+    dbg("building model")
+    cfg = {"llama3_8b": LLAMA3_8B, "llama3_70b": LLAMA3_70B, "debug": LLAMA_DEBUG}[
+        args.model
+    ]
+    seq = min(args.seq, cfg.max_seq_len)
+    torch.manual_seed(1234)
+    dtype = torch.bfloat16
+    model = Llama(cfg, dtype=dtype, checkpoint_activations=args.model != "debug")
+    model = model.to(device)
+
+    # ---- fault-tolerance control plane ------------------------------------
+    dbg("model built")
+    manager = None
+    if not args.no_ft:
+        from torchft_amd._ftcore import LighthouseServer
+        from torchft_amd.manager import Manager
+        from torchft_amd.process_group import ProcessGroupRCCL
+
+        lighthouse = None
+        if rank == 0:
+            lighthouse = LighthouseServer(
+                bind="0.0.0.0:0", min_replicas=replicas, join_timeout_ms=5000
+            )
+            boot_store.set("lighthouse_addr", lighthouse.address())
+        lighthouse_addr = boot_store.get("lighthouse_addr").decode()
+
+        ft_pg = (
+            ProcessGroupRCCL(timeout=timedelta(seconds=120))
+            if use_cuda
+            else ProcessGroupGloo(timeout=timedelta(seconds=120))
+        )
+        manager = Manager(
+            pg=ft_pg,
+            load_state_dict=model.load_state_dict,
+            state_dict=model.state_dict,
+            min_replica_size=replicas,
+            rank=group_rank,
+            world_size=shards,
+            store_addr=master_addr,
+            store_port=mgr_port,
+            lighthouse_addr=lighthouse_addr,
+            replica_id=f"bench{group}",
+            hostname=master_addr,
+            timeout=timedelta(seconds=120),
+            quorum_timeout=timedelta(seconds=120),
+            connect_timeout=timedelta(seconds=60),
+            # identical random init on every replica; no step-0 state transfer
+            init_sync=False,
+        )
+
+    dbg("manager ready")
+    # ---- parallelism wiring ------------------------------------------------
+    # Route forward to forward_loss BEFORE wrapping so DDP's reducer and
+    # FSDP's pre/post-forward hooks fire through __call__.
+    model.forward = model.forward_loss  # type: ignore[assignment]
+    ddp_model = model
+    if shards > 1:
+        from torch.distributed.fsdp import FSDPModule, fully_shard
+
+        for layer in model.layers:
+            fully_shard(layer)
+        fully_shard(model)
+
+        if manager is not None:
+
+            def all_reduce_hook(output: torch.Tensor) -> None:
+                manager.allreduce(output).wait()
+
+            def apply_hook(m: torch.nn.Module) -> None:
+                if isinstance(m, FSDPModule):
+                    m.set_all_reduce_hook(all_reduce_hook)
+
+            model.apply(apply_hook)
+        ddp_model = model
+    elif manager is not None and world > 1:
+        from torchft_amd.ddp import DistributedDataParallel
+
+        ddp_model = DistributedDataParallel(manager, model)
+
+    # ---- optimizer ---------------------------------------------------------
+    if shards > 1:
+        base_opt = torch.optim.AdamW(
+            model.parameters(), lr=3e-4, betas=(0.9, 0.95), weight_decay=0.1,
+            foreach=True,
+        )
+    else:
+        from torchft_amd.ops import FusedAdamW
+
+        base_opt = FusedAdamW(model.parameters(), lr=3e-4, betas=(0.9, 0.95),
+                              weight_decay=0.1)
+
+    if manager is not None:
+        from torchft_amd.optim import OptimizerWrapper
+
+        opt = OptimizerWrapper(manager, base_opt)
+    else:
+        opt = base_opt
+
+    # ---- synthetic data -----------------------------------------------------
+    gen = torch.Generator(device="cpu").manual_seed(42 + rank)
+    def make_batch():
+        toks = torch.randint(0, cfg.vocab_size, (args.batch, seq + 1), generator=gen)
+        x = toks[:, :-1].to(device, non_blocking=True)
+        y = toks[:, 1:].to(device, non_blocking=True)
+        return x, y
+
+    def one_step() -> float:
+        x, y = make_batch()
+        opt.zero_grad()  # OptimizerWrapper: starts the async quorum
+        loss = ddp_model(x, y)
+        loss.backward()
+        opt.step()  # OptimizerWrapper: should_commit barrier gates the step
+        return float(loss.detach())
+
+    # ---- run ----------------------------------------------------------------
+    dbg("wiring done, starting warmup")
+    for i in range(args.warmup):
+        loss = one_step()
+        if rank == 0:
+            print(f"[warmup {i}] loss={loss:.4f}", file=sys.stderr, flush=True)
+
+    global_barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        loss = one_step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    global_barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max elapsed over ranks (gloo allreduce MAX)
+    from torch.distributed import ReduceOp as RO
+    from torch.distributed.distributed_c10d import AllreduceOptions
+
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if world > 1:
+        opts = AllreduceOptions()
+        opts.reduceOp = RO.MAX
+        barrier_pg.allreduce([t], opts).wait()
+    elapsed = float(t[0])
+
+    tokens_total = args.batch * seq * world * args.steps
+    tok_s = tokens_total / elapsed
+    ms_per_step = elapsed / args.steps * 1000
+
+    if rank == 0:
+        result = {
+            "metric": "goodput_tokens_per_sec",
+            "value": tok_s,
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * world,
+                "seq_len": seq,
+                "parallelism": f"ft-hsdp{replicas}x{shards}" if shards > 1 else f"ft-dp{world}",
+                "fault_tolerance": manager is not None,
+                "loss": loss,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if manager is not None:
+        manager.shutdown(wait=False)
+        if rank == 0 and lighthouse is not None:
+            lighthouse.shutdown()
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

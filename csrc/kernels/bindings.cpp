@@ -1,0 +1,255 @@
+// torch-extension bindings for the CDNA4 kernels (fp8 quantized-collective
+// trio, fused model ops, fused AdamW). Tensor-level API consumed by
+// torchft_amd.ops / torchft_amd.quantization.
+#include <ATen/cuda/CUDAContext.h>
+#include <torch/extension.h>
+
+#include <cmath>
+
+#include <vector>
+
+#include "kernels.h"
+
+namespace tft = torchft_amd;
+
+namespace {
+
+constexpr int64_t kQBlock = 2048;
+
+struct PackGeom {
+  int64_t total_blocks = 0;
+  int64_t padded_blocks = 0;
+  int64_t blocks_per_rank = 0;
+  int64_t slice_bytes = 0;
+};
+
+PackGeom pack_geometry(const std::vector<at::Tensor>& tensors, int64_t world) {
+  PackGeom g;
+  for (auto& t : tensors) g.total_blocks += (t.numel() + kQBlock - 1) / kQBlock;
+  g.padded_blocks = ((g.total_blocks + world - 1) / world) * world;
+  g.blocks_per_rank = g.padded_blocks / world;
+  g.slice_bytes = g.blocks_per_rank * (4 + kQBlock);
+  return g;
+}
+
+// Builds the device-side metadata arrays (ptrs, block prefix, numels).
+struct DeviceMeta {
+  at::Tensor ptrs, prefix, numels;
+};
+
+DeviceMeta build_meta(const std::vector<at::Tensor>& tensors) {
+  const int64_t n = (int64_t)tensors.size();
+  auto opts = at::TensorOptions().dtype(at::kLong).pinned_memory(true);
+  at::Tensor host = at::empty({3 * n + 1}, opts);
+  int64_t* h = host.data_ptr<int64_t>();
+  int64_t* ptrs = h;
+  int64_t* numels = h + n;
+  int64_t* prefix = h + 2 * n;  // n+1 entries
+  int64_t acc = 0;
+  for (int64_t i = 0; i < n; i++) {
+    auto& t = tensors[i];
+    TORCH_CHECK(t.is_contiguous(), "fp8 pack requires contiguous tensors");
+    TORCH_CHECK(t.is_cuda(), "fp8 pack requires device tensors");
+    ptrs[i] = (int64_t)t.data_ptr();
+    numels[i] = t.numel();
+    prefix[i] = acc;
+    acc += (t.numel() + kQBlock - 1) / kQBlock;
+  }
+  prefix[n] = acc;
+  at::Tensor dev = host.to(tensors[0].device(), /*non_blocking=*/true);
+  DeviceMeta m;
+  m.ptrs = dev.narrow(0, 0, n);
+  m.numels = dev.narrow(0, n, n);
+  m.prefix = dev.narrow(0, 2 * n, n + 1);
+  return m;
+}
+
+}  // namespace
+
+// ---- fp8 quantized-collective support --------------------------------------
+
+int64_t fp8_pack_bytes(const std::vector<at::Tensor>& tensors, int64_t world) {
+  auto g = pack_geometry(tensors, world);
+  return g.slice_bytes * world;
+}
+
+int64_t fp8_slice_bytes(const std::vector<at::Tensor>& tensors, int64_t world) {
+  return pack_geometry(tensors, world).slice_bytes;
+}
+
+void fp8_quantize(const std::vector<at::Tensor>& tensors, at::Tensor pack,
+                  int64_t world) {
+  TORCH_CHECK(!tensors.empty(), "need at least one tensor");
+  auto g = pack_geometry(tensors, world);
+  TORCH_CHECK(pack.numel() >= g.slice_bytes * world, "pack buffer too small");
+  auto meta = build_meta(tensors);
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  auto st = tensors[0].scalar_type();
+  for (auto& t : tensors) TORCH_CHECK(t.scalar_type() == st, "mixed dtypes");
+  const int64_t* ptrs = meta.ptrs.data_ptr<int64_t>();
+  const int64_t* prefix = meta.prefix.data_ptr<int64_t>();
+  const int64_t* numels = meta.numels.data_ptr<int64_t>();
+  uint8_t* p = pack.data_ptr<uint8_t>();
+  int code = st == at::kBFloat16 ? 0 : st == at::kHalf ? 1 : st == at::kFloat ? 2 : -1;
+  TORCH_CHECK(code >= 0, "unsupported dtype for fp8 quantize: ", st);
+  tft::launch_quantize_dtype(code, ptrs, prefix, numels, (int)tensors.size(),
+                             g.total_blocks, g.padded_blocks, g.blocks_per_rank,
+                             g.slice_bytes, p, (tft_stream)stream);
+}
+
+void fp8_dequantize(const std::vector<at::Tensor>& tensors, at::Tensor pack,
+                    int64_t world) {
+  TORCH_CHECK(!tensors.empty(), "need at least one tensor");
+  auto g = pack_geometry(tensors, world);
+  auto meta = build_meta(tensors);
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  auto st = tensors[0].scalar_type();
+  const int64_t* ptrs = meta.ptrs.data_ptr<int64_t>();
+  const int64_t* prefix = meta.prefix.data_ptr<int64_t>();
+  const int64_t* numels = meta.numels.data_ptr<int64_t>();
+  const uint8_t* p = pack.data_ptr<uint8_t>();
+  int code = st == at::kBFloat16 ? 0 : st == at::kHalf ? 1 : st == at::kFloat ? 2 : -1;
+  TORCH_CHECK(code >= 0, "unsupported dtype for fp8 dequantize: ", st);
+  tft::launch_dequantize_dtype(code, ptrs, prefix, numels, (int)tensors.size(),
+                               g.total_blocks, g.padded_blocks, g.blocks_per_rank,
+                               g.slice_bytes, p, (tft_stream)stream);
+}
+
+void fp8_reduce(at::Tensor recv, at::Tensor out, int64_t world, bool avg) {
+  TORCH_CHECK(recv.is_cuda() && out.is_cuda());
+  TORCH_CHECK(recv.numel() == out.numel() * world, "recv must hold world slices");
+  const int64_t slice_bytes = out.numel();
+  const int64_t blocks_per_rank = slice_bytes / (4 + kQBlock);
+  TORCH_CHECK(blocks_per_rank * (4 + kQBlock) == slice_bytes, "bad slice size");
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  tft::launch_reduce(recv.data_ptr<uint8_t>(), out.data_ptr<uint8_t>(), (int)world,
+                     blocks_per_rank, slice_bytes, avg, (tft_stream)stream);
+}
+
+// ---- fused model ops --------------------------------------------------------
+
+std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 && x.is_contiguous());
+  TORCH_CHECK(w.is_cuda() && w.scalar_type() == at::kBFloat16 && w.is_contiguous());
+  const int H = (int)x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "hidden dim must be a multiple of 8");
+  const int64_t rows = x.numel() / H;
+  auto y = at::empty_like(x);
+  auto invrms = at::empty({rows}, x.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  tft::launch_rmsnorm_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                          invrms.data_ptr<float>(), rows, H, (float)eps,
+                          (tft_stream)stream);
+  return {y, invrms};
+}
+
+std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
+                                    at::Tensor invrms) {
+  const int H = (int)x.size(-1);
+  const int64_t rows = x.numel() / H;
+  TORCH_CHECK(H * sizeof(float) <= 160 * 1024, "H too large for LDS dw buffer");
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto dyc = dy.contiguous();
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  tft::launch_rmsnorm_bwd(dyc.data_ptr(), x.data_ptr(), w.data_ptr(),
+                          invrms.data_ptr<float>(), dx.data_ptr(),
+                          dw.data_ptr<float>(), rows, H, (tft_stream)stream);
+  return {dx, dw};
+}
+
+at::Tensor rope_apply(at::Tensor x, at::Tensor cos_tab, at::Tensor sin_tab,
+                      bool backward) {
+  // x: [B, S, Hh, D]
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 && x.dim() == 4);
+  auto xc = x.contiguous();
+  const int S = (int)x.size(1);
+  const int Hh = (int)x.size(2);
+  const int D = (int)x.size(3);
+  TORCH_CHECK(D % 8 == 0, "head dim must be a multiple of 8");
+  TORCH_CHECK(cos_tab.size(0) >= S && cos_tab.size(1) == D / 2, "cos table shape");
+  auto out = at::empty_like(xc);
+  const int64_t total_pairs = xc.numel() / 2;
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  tft::launch_rope(xc.data_ptr(), out.data_ptr(), cos_tab.data_ptr<float>(),
+                   sin_tab.data_ptr<float>(), total_pairs, S, Hh, D, backward,
+                   (tft_stream)stream);
+  return out;
+}
+
+at::Tensor swiglu_fwd(at::Tensor a, at::Tensor b) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(a.numel() % 8 == 0, "numel must be a multiple of 8");
+  auto ac = a.contiguous();
+  auto bc = b.contiguous();
+  auto out = at::empty_like(ac);
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  tft::launch_swiglu_fwd(ac.data_ptr(), bc.data_ptr(), out.data_ptr(), ac.numel(),
+                         (tft_stream)stream);
+  return out;
+}
+
+std::vector<at::Tensor> swiglu_bwd(at::Tensor dy, at::Tensor a, at::Tensor b) {
+  auto dyc = dy.contiguous();
+  auto ac = a.contiguous();
+  auto bc = b.contiguous();
+  auto da = at::empty_like(ac);
+  auto db = at::empty_like(bc);
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  tft::launch_swiglu_bwd(dyc.data_ptr(), ac.data_ptr(), bc.data_ptr(),
+                         da.data_ptr(), db.data_ptr(), ac.numel(),
+                         (tft_stream)stream);
+  return {da, db};
+}
+
+// ---- fused AdamW ------------------------------------------------------------
+
+void adamw_step(const std::vector<at::Tensor>& params,
+                const std::vector<at::Tensor>& grads,
+                const std::vector<at::Tensor>& exp_avgs,
+                const std::vector<at::Tensor>& exp_avg_sqs, double lr,
+                double beta1, double beta2, double eps, double weight_decay,
+                int64_t step) {
+  const int64_t n = (int64_t)params.size();
+  TORCH_CHECK(n > 0);
+  auto opts = at::TensorOptions().dtype(at::kLong).pinned_memory(true);
+  at::Tensor host = at::empty({6 * n + 1}, opts);
+  int64_t* h = host.data_ptr<int64_t>();
+  int64_t acc = 0;
+  for (int64_t i = 0; i < n; i++) {
+    TORCH_CHECK(params[i].is_contiguous() && grads[i].is_contiguous());
+    TORCH_CHECK(params[i].scalar_type() == at::kBFloat16, "params must be bf16");
+    TORCH_CHECK(exp_avgs[i].scalar_type() == at::kFloat, "states must be fp32");
+    h[i] = (int64_t)params[i].data_ptr();
+    h[n + i] = (int64_t)grads[i].data_ptr();
+    h[2 * n + i] = (int64_t)exp_avgs[i].data_ptr();
+    h[3 * n + i] = (int64_t)exp_avg_sqs[i].data_ptr();
+    h[4 * n + i] = params[i].numel();
+    h[5 * n + i] = acc;
+    acc += (params[i].numel() + kQBlock - 1) / kQBlock;
+  }
+  h[6 * n] = acc;
+  at::Tensor dev = host.to(params[0].device(), /*non_blocking=*/true);
+  const int64_t* d = dev.data_ptr<int64_t>();
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  const float bc1 = 1.0f - std::pow((float)beta1, (float)step);
+  const float bc2 = 1.0f - std::pow((float)beta2, (float)step);
+  tft::launch_adamw(d, d + n, d + 2 * n, d + 3 * n, d + 4 * n, d + 5 * n, (int)n,
+                    acc, (float)lr, (float)beta1, (float)beta2, (float)eps,
+                    (float)weight_decay, bc1, bc2, (tft_stream)stream);
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "torchft_amd CDNA4 (gfx950) HIP kernels";
+  m.def("fp8_pack_bytes", &fp8_pack_bytes);
+  m.def("fp8_slice_bytes", &fp8_slice_bytes);
+  m.def("fp8_quantize", &fp8_quantize);
+  m.def("fp8_dequantize", &fp8_dequantize);
+  m.def("fp8_reduce", &fp8_reduce);
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rope_apply", &rope_apply);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("adamw_step", &adamw_step);
+}

@@ -1,0 +1,135 @@
+// Multi-tensor fused AdamW step — CDNA4 (gfx950) native.
+//
+// One kernel launch updates every parameter tensor: bf16 params/grads,
+// fp32 exp_avg/exp_avg_sq. The block→tensor mapping reuses the fp8-quant
+// scheme (2048-element blocks, per-tensor padding, wave-uniform binary
+// search) so the launch is a single uniform grid regardless of the
+// parameter-shape zoo. HBM-bound: 4 reads + 3 writes per element, all
+// 16 B/lane vectorized.
+//
+// Serves as the DiLoCo fused outer-optimizer step and the bench's inner
+// optimizer (reference analogue: torchft delegates to torch.optim).
+
+#include <hip/hip_bf16.h>
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+namespace torchft_amd {
+
+using bf16 = __hip_bfloat16;
+using bf16x2 = __hip_bfloat162;
+
+#define ABLOCK 2048
+#define ATHREADS 256
+#define AEPT 8  // elements per thread
+
+__device__ inline int find_tensor_a(const int64_t* block_prefix, int n, int64_t b) {
+  int lo = 0, hi = n;
+  while (hi - lo > 1) {
+    int mid = (lo + hi) >> 1;
+    if (block_prefix[mid] <= b)
+      lo = mid;
+    else
+      hi = mid;
+  }
+  return lo;
+}
+
+__global__ void adamw_kernel(const int64_t* __restrict__ param_ptrs,
+                             const int64_t* __restrict__ grad_ptrs,
+                             const int64_t* __restrict__ m_ptrs,
+                             const int64_t* __restrict__ v_ptrs,
+                             const int64_t* __restrict__ numels,
+                             const int64_t* __restrict__ block_prefix,
+                             int n_tensors, float lr, float beta1, float beta2,
+                             float eps, float weight_decay, float bc1, float bc2) {
+  const int64_t b = blockIdx.x;
+  const int t = find_tensor_a(block_prefix, n_tensors, b);
+  const int64_t off = (b - block_prefix[t]) * ABLOCK + threadIdx.x * AEPT;
+  const int64_t numel = numels[t];
+  if (off >= numel) return;
+
+  bf16* p = reinterpret_cast<bf16*>(param_ptrs[t]) + off;
+  const bf16* g = reinterpret_cast<const bf16*>(grad_ptrs[t]) + off;
+  float* m = reinterpret_cast<float*>(m_ptrs[t]) + off;
+  float* v = reinterpret_cast<float*>(v_ptrs[t]) + off;
+
+  const bool full = off + AEPT <= numel;
+  const int cnt = full ? AEPT : (int)(numel - off);
+
+  float pv[AEPT], gv[AEPT], mv[AEPT], vv[AEPT];
+  if (full) {
+    const ushort4 rp = *reinterpret_cast<const ushort4*>(p);
+    const ushort4 rg = *reinterpret_cast<const ushort4*>(g);
+    const bf16x2* hp = reinterpret_cast<const bf16x2*>(&rp);
+    const bf16x2* hg = reinterpret_cast<const bf16x2*>(&rg);
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+      float2 fp = __bfloat1622float2(hp[i]);
+      float2 fg = __bfloat1622float2(hg[i]);
+      pv[2 * i] = fp.x; pv[2 * i + 1] = fp.y;
+      gv[2 * i] = fg.x; gv[2 * i + 1] = fg.y;
+    }
+    const float4 m0 = *reinterpret_cast<const float4*>(m);
+    const float4 m1 = *reinterpret_cast<const float4*>(m + 4);
+    const float4 v0 = *reinterpret_cast<const float4*>(v);
+    const float4 v1 = *reinterpret_cast<const float4*>(v + 4);
+    mv[0] = m0.x; mv[1] = m0.y; mv[2] = m0.z; mv[3] = m0.w;
+    mv[4] = m1.x; mv[5] = m1.y; mv[6] = m1.z; mv[7] = m1.w;
+    vv[0] = v0.x; vv[1] = v0.y; vv[2] = v0.z; vv[3] = v0.w;
+    vv[4] = v1.x; vv[5] = v1.y; vv[6] = v1.z; vv[7] = v1.w;
+  } else {
+    for (int i = 0; i < cnt; i++) {
+      pv[i] = __bfloat162float(p[i]);
+      gv[i] = __bfloat162float(g[i]);
+      mv[i] = m[i];
+      vv[i] = v[i];
+    }
+  }
+
+#pragma unroll
+  for (int i = 0; i < AEPT; i++) {
+    if (!full && i >= cnt) break;
+    mv[i] = beta1 * mv[i] + (1.f - beta1) * gv[i];
+    vv[i] = beta2 * vv[i] + (1.f - beta2) * gv[i] * gv[i];
+    const float mhat = mv[i] / bc1;
+    const float vhat = vv[i] / bc2;
+    // decoupled weight decay
+    pv[i] -= lr * (mhat / (sqrtf(vhat) + eps) + weight_decay * pv[i]);
+  }
+
+  if (full) {
+    ushort4 rp;
+    bf16x2* hp = reinterpret_cast<bf16x2*>(&rp);
+#pragma unroll
+    for (int i = 0; i < 4; i++)
+      hp[i] = __float22bfloat162_rn(make_float2(pv[2 * i], pv[2 * i + 1]));
+    *reinterpret_cast<ushort4*>(p) = rp;
+    *reinterpret_cast<float4*>(m) = make_float4(mv[0], mv[1], mv[2], mv[3]);
+    *reinterpret_cast<float4*>(m + 4) = make_float4(mv[4], mv[5], mv[6], mv[7]);
+    *reinterpret_cast<float4*>(v) = make_float4(vv[0], vv[1], vv[2], vv[3]);
+    *reinterpret_cast<float4*>(v + 4) = make_float4(vv[4], vv[5], vv[6], vv[7]);
+  } else {
+    for (int i = 0; i < cnt; i++) {
+      p[i] = __float2bfloat16(pv[i]);
+      m[i] = mv[i];
+      v[i] = vv[i];
+    }
+  }
+}
+
+void launch_adamw(const int64_t* param_ptrs, const int64_t* grad_ptrs,
+                  const int64_t* m_ptrs, const int64_t* v_ptrs,
+                  const int64_t* numels, const int64_t* block_prefix,
+                  int n_tensors, int64_t total_blocks, float lr, float beta1,
+                  float beta2, float eps, float weight_decay,
+                  float bias_correction1, float bias_correction2,
+                  hipStream_t stream) {
+  hipLaunchKernelGGL(adamw_kernel, dim3((uint32_t)total_blocks), dim3(ATHREADS), 0,
+                     stream, param_ptrs, grad_ptrs, m_ptrs, v_ptrs, numels,
+                     block_prefix, n_tensors, lr, beta1, beta2, eps, weight_decay,
+                     bias_correction1, bias_correction2);
+}
+
+}  // namespace torchft_amd

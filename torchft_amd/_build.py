@@ -64,13 +64,14 @@ def build_hip_kernels(force: bool = False) -> Path | None:
     """Compile the gfx950 HIP kernel extension via torch.utils.cpp_extension.
 
     Cross-compiles on machines with no GPU (hipcc targets gfx950 regardless).
+    The built .so is copied in-tree (torchft_amd/_hip_kernels.so) so the repo
+    snapshot shipped to GPU boxes carries it.
     """
     src_dir = CSRC / "kernels"
-    sources = sorted(src_dir.glob("*.hip")) + sorted(src_dir.glob("*.cpp"))
-    if not sources:
-        return None
-    out = PKG / f"_hip_kernels{EXT_SUFFIX}"
-    if not force and _newer(out, sources):
+    sources = sorted(src_dir.glob("*.hip")) + [src_dir / "bindings.cpp"]
+    headers = [src_dir / "kernels.h"]
+    out = PKG / "_hip_kernels.so"
+    if not force and _newer(out, sources + headers):
         return out
 
     os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
@@ -79,24 +80,18 @@ def build_hip_kernels(force: bool = False) -> Path | None:
 
     build_dir = REPO / "build" / "hip_kernels"
     build_dir.mkdir(parents=True, exist_ok=True)
-    mod = load(
+    load(
         name="_hip_kernels",
         sources=[str(s) for s in sources],
         extra_cflags=["-O3"],
-        extra_cuda_cflags=["-O3", "--offload-arch=gfx950"],
+        extra_cuda_cflags=["-O3"],
         build_directory=str(build_dir),
         verbose=True,
-        is_python_module=False,
-        is_standalone=False,
-        keep_intermediates=True,
     )
-    # torch's load() with is_python_module=False loads into the process; we
-    # still copy the built .so in-tree so it snapshots to GPU boxes.
-    built = build_dir / f"_hip_kernels.so"
-    if built.exists():
-        import shutil
+    built = build_dir / "_hip_kernels.so"
+    import shutil
 
-        shutil.copy2(built, out)
+    shutil.copy2(built, out)
     return out
 
 

@@ -1,0 +1,168 @@
+"""Llama-3 model family for the flagship fault-tolerant training benchmark.
+
+New scope relative to the reference (torchft delegates the model stack to
+torchtitan; BASELINE.json names Llama-3 8B/70B as the benchmark configs).
+MI355X-first: bf16 parameters, fused CDNA4 RMSNorm/RoPE/SwiGLU kernels
+(torchft_amd.ops), SDPA flash attention with GQA, chunked cross-entropy so
+the 128k-vocab logits never materialize at once, and optional per-block
+activation checkpointing sized for 288 GB HBM3E.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from torchft_amd.ops import RMSNorm, rope, rope_tables, swiglu
+
+
+@dataclass
+class LlamaConfig:
+    dim: int = 4096
+    n_layers: int = 32
+    n_heads: int = 32
+    n_kv_heads: int = 8
+    ffn_hidden: int = 14336
+    vocab_size: int = 128256
+    max_seq_len: int = 8192
+    rope_theta: float = 500000.0
+    norm_eps: float = 1e-5
+
+    @property
+    def head_dim(self) -> int:
+        return self.dim // self.n_heads
+
+
+LLAMA3_8B = LlamaConfig()
+LLAMA3_70B = LlamaConfig(
+    dim=8192, n_layers=80, n_heads=64, n_kv_heads=8, ffn_hidden=28672
+)
+LLAMA_DEBUG = LlamaConfig(
+    dim=256, n_layers=2, n_heads=4, n_kv_heads=2, ffn_hidden=512,
+    vocab_size=512, max_seq_len=512,
+)
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg: LlamaConfig, dtype: torch.dtype) -> None:
+        super().__init__()
+        self.cfg = cfg
+        d, hd = cfg.dim, cfg.head_dim
+        self.wq = nn.Linear(d, cfg.n_heads * hd, bias=False, dtype=dtype)
+        self.wk = nn.Linear(d, cfg.n_kv_heads * hd, bias=False, dtype=dtype)
+        self.wv = nn.Linear(d, cfg.n_kv_heads * hd, bias=False, dtype=dtype)
+        self.wo = nn.Linear(cfg.n_heads * hd, d, bias=False, dtype=dtype)
+
+    def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+        B, S, _ = x.shape
+        cfg = self.cfg
+        q = self.wq(x).view(B, S, cfg.n_heads, cfg.head_dim)
+        k = self.wk(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
+        v = self.wv(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
+        q = rope(q, cos, sin)
+        k = rope(k, cos, sin)
+        # SDPA wants [B, H, S, D]
+        q, k, v = (t.transpose(1, 2) for t in (q, k, v))
+        out = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)
+        out = out.transpose(1, 2).reshape(B, S, -1)
+        return self.wo(out)
+
+
+class MLP(nn.Module):
+    def __init__(self, cfg: LlamaConfig, dtype: torch.dtype) -> None:
+        super().__init__()
+        self.w1 = nn.Linear(cfg.dim, cfg.ffn_hidden, bias=False, dtype=dtype)  # gate
+        self.w3 = nn.Linear(cfg.dim, cfg.ffn_hidden, bias=False, dtype=dtype)  # up
+        self.w2 = nn.Linear(cfg.ffn_hidden, cfg.dim, bias=False, dtype=dtype)  # down
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.w2(swiglu(self.w1(x), self.w3(x)))
+
+
+class Block(nn.Module):
+    def __init__(self, cfg: LlamaConfig, dtype: torch.dtype) -> None:
+        super().__init__()
+        self.attn_norm = RMSNorm(cfg.dim, cfg.norm_eps, dtype=dtype)
+        self.attn = Attention(cfg, dtype)
+        self.mlp_norm = RMSNorm(cfg.dim, cfg.norm_eps, dtype=dtype)
+        self.mlp = MLP(cfg, dtype)
+
+    def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+        x = x + self.attn(self.attn_norm(x), cos, sin)
+        x = x + self.mlp(self.mlp_norm(x))
+        return x
+
+
+class Llama(nn.Module):
+    def __init__(
+        self,
+        cfg: LlamaConfig,
+        dtype: torch.dtype = torch.bfloat16,
+        checkpoint_activations: bool = True,
+    ) -> None:
+        super().__init__()
+        self.cfg = cfg
+        self.checkpoint_activations = checkpoint_activations
+        self.tok_embeddings = nn.Embedding(cfg.vocab_size, cfg.dim, dtype=dtype)
+        self.layers = nn.ModuleList(Block(cfg, dtype) for _ in range(cfg.n_layers))
+        self.norm = RMSNorm(cfg.dim, cfg.norm_eps, dtype=dtype)
+        self.output = nn.Linear(cfg.dim, cfg.vocab_size, bias=False, dtype=dtype)
+
+        cos, sin = rope_tables(cfg.max_seq_len, cfg.head_dim, cfg.rope_theta)
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+
+        self._init_weights()
+
+    def _init_weights(self) -> None:
+        std = 0.02
+        for m in self.modules():
+            if isinstance(m, nn.Linear):
+                nn.init.normal_(m.weight, mean=0.0, std=std)
+            elif isinstance(m, nn.Embedding):
+                nn.init.normal_(m.weight, mean=0.0, std=std)
+
+    def forward_hidden(self, tokens: torch.Tensor) -> torch.Tensor:
+        """Token ids [B, S] → final hidden states [B, S, dim]."""
+        x = self.tok_embeddings(tokens)
+        cos, sin = self.rope_cos, self.rope_sin
+        for layer in self.layers:
+            if self.checkpoint_activations and self.training:
+                x = torch.utils.checkpoint.checkpoint(
+                    layer, x, cos, sin, use_reentrant=False
+                )
+            else:
+                x = layer(x, cos, sin)
+        return self.norm(x)
+
+    def forward(self, tokens: torch.Tensor) -> torch.Tensor:
+        """Full logits — use forward_loss for training (chunked CE)."""
+        return self.output(self.forward_hidden(tokens))
+
+    def forward_loss(
+        self,
+        tokens: torch.Tensor,
+        targets: torch.Tensor,
+        chunk_rows: int = 4096,
+    ) -> torch.Tensor:
+        """Mean cross-entropy with chunked logits: the [B*S, vocab] logits
+        matrix (2 GB+ at 128k vocab) is computed ``chunk_rows`` rows at a
+        time so peak memory stays bounded."""
+        h = self.forward_hidden(tokens)
+        h = h.reshape(-1, self.cfg.dim)
+        t = targets.reshape(-1)
+        n = h.shape[0]
+        losses = []
+        for i in range(0, n, chunk_rows):
+            logits = self.output(h[i : i + chunk_rows])
+            losses.append(
+                F.cross_entropy(logits.float(), t[i : i + chunk_rows], reduction="sum")
+            )
+        return torch.stack(losses).sum() / n
+
+    def num_params(self) -> int:
+        return sum(p.numel() for p in self.parameters())

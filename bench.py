@@ -299,7 +299,7 @@ def main() -> None:
                 "seq_len": seq,
                 "parallelism": f"ft-hsdp{replicas}x{shards}" if shards > 1 else f"ft-dp{world}",
                 "fault_tolerance": manager is not None,
-                "loss": loss,
+                "loss": loss if loss == loss else "nan",  # keep JSON strict
             },
         }
         print(json.dumps(result), flush=True)

@@ -49,7 +49,7 @@ template <>
 struct VecIO<__hip_bfloat16> {
   // 8 bf16 = 16 bytes
   static __device__ inline void load8(const __hip_bfloat16* p, float (&v)[8]) {
-    const ushort4 raw = *reinterpret_cast<const ushort4*>(p);
+    const uint4 raw = *reinterpret_cast<const uint4*>(p);  // 16 bytes
     const __hip_bfloat162* h = reinterpret_cast<const __hip_bfloat162*>(&raw);
 #pragma unroll
     for (int i = 0; i < 4; i++) {
@@ -59,20 +59,20 @@ struct VecIO<__hip_bfloat16> {
     }
   }
   static __device__ inline void store8(__hip_bfloat16* p, const float (&v)[8]) {
-    ushort4 raw;
+    uint4 raw;
     __hip_bfloat162* h = reinterpret_cast<__hip_bfloat162*>(&raw);
 #pragma unroll
     for (int i = 0; i < 4; i++) {
       h[i] = __float22bfloat162_rn(make_float2(v[2 * i], v[2 * i + 1]));
     }
-    *reinterpret_cast<ushort4*>(p) = raw;
+    *reinterpret_cast<uint4*>(p) = raw;
   }
 };
 
 template <>
 struct VecIO<__half> {
   static __device__ inline void load8(const __half* p, float (&v)[8]) {
-    const ushort4 raw = *reinterpret_cast<const ushort4*>(p);
+    const uint4 raw = *reinterpret_cast<const uint4*>(p);  // 16 bytes
     const __half2* h = reinterpret_cast<const __half2*>(&raw);
 #pragma unroll
     for (int i = 0; i < 4; i++) {
@@ -82,13 +82,13 @@ struct VecIO<__half> {
     }
   }
   static __device__ inline void store8(__half* p, const float (&v)[8]) {
-    ushort4 raw;
+    uint4 raw;
     __half2* h = reinterpret_cast<__half2*>(&raw);
 #pragma unroll
     for (int i = 0; i < 4; i++) {
       h[i] = __float22half2_rn(make_float2(v[2 * i], v[2 * i + 1]));
     }
-    *reinterpret_cast<ushort4*>(p) = raw;
+    *reinterpret_cast<uint4*>(p) = raw;
   }
 };
 

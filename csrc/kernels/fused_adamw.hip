@@ -60,8 +60,8 @@ __global__ void adamw_kernel(const int64_t* __restrict__ param_ptrs,
 
   float pv[AEPT], gv[AEPT], mv[AEPT], vv[AEPT];
   if (full) {
-    const ushort4 rp = *reinterpret_cast<const ushort4*>(p);
-    const ushort4 rg = *reinterpret_cast<const ushort4*>(g);
+    const uint4 rp = *reinterpret_cast<const uint4*>(p);  // 8 bf16 = 16 B
+    const uint4 rg = *reinterpret_cast<const uint4*>(g);
     const bf16x2* hp = reinterpret_cast<const bf16x2*>(&rp);
     const bf16x2* hg = reinterpret_cast<const bf16x2*>(&rg);
 #pragma unroll
@@ -100,12 +100,12 @@ __global__ void adamw_kernel(const int64_t* __restrict__ param_ptrs,
   }
 
   if (full) {
-    ushort4 rp;
+    uint4 rp;
     bf16x2* hp = reinterpret_cast<bf16x2*>(&rp);
 #pragma unroll
     for (int i = 0; i < 4; i++)
       hp[i] = __float22bfloat162_rn(make_float2(pv[2 * i], pv[2 * i + 1]));
-    *reinterpret_cast<ushort4*>(p) = rp;
+    *reinterpret_cast<uint4*>(p) = rp;
     *reinterpret_cast<float4*>(m) = make_float4(mv[0], mv[1], mv[2], mv[3]);
     *reinterpret_cast<float4*>(m + 4) = make_float4(mv[4], mv[5], mv[6], mv[7]);
     *reinterpret_cast<float4*>(v) = make_float4(vv[0], vv[1], vv[2], vv[3]);

@@ -22,7 +22,8 @@ using bf16x2 = __hip_bfloat162;
 // ---------------------------------------------------------------- helpers
 
 __device__ inline void load_bf16x8(const bf16* p, float (&v)[8]) {
-  const ushort4 raw = *reinterpret_cast<const ushort4*>(p);
+  // 8 bf16 = 16 bytes = one uint4 (ushort4 is only 8 bytes!)
+  const uint4 raw = *reinterpret_cast<const uint4*>(p);
   const bf16x2* h = reinterpret_cast<const bf16x2*>(&raw);
 #pragma unroll
   for (int i = 0; i < 4; i++) {
@@ -33,11 +34,11 @@ __device__ inline void load_bf16x8(const bf16* p, float (&v)[8]) {
 }
 
 __device__ inline void store_bf16x8(bf16* p, const float (&v)[8]) {
-  ushort4 raw;
+  uint4 raw;
   bf16x2* h = reinterpret_cast<bf16x2*>(&raw);
 #pragma unroll
   for (int i = 0; i < 4; i++) h[i] = __float22bfloat162_rn(make_float2(v[2 * i], v[2 * i + 1]));
-  *reinterpret_cast<ushort4*>(p) = raw;
+  *reinterpret_cast<uint4*>(p) = raw;
 }
 
 template <int THREADS>

@@ -116,21 +116,24 @@ class TestHTTPTransport:
             src.shutdown()
             dst.shutdown()
 
-    def test_disallow_blocks_serving(self):
+    def test_snapshot_survives_disallow(self):
+        # The staged checkpoint is a snapshot, so (unlike the reference's
+        # live-serving transport) it stays fetchable after training resumes;
+        # this removes the slow-fetcher vs next-step race.
         src = HTTPTransport(timeout=timedelta(seconds=10))
         dst = HTTPTransport(timeout=timedelta(seconds=10))
         try:
+            t = torch.ones(4)
             src.send_checkpoint(
-                [1], step=3, state_dict={"x": 1}, timeout=timedelta(seconds=10)
+                [1], step=3, state_dict={"x": t}, timeout=timedelta(seconds=10)
             )
+            t.mul_(100.0)  # training resumed and mutated the weights
             src.disallow_checkpoint()
-            with pytest.raises(Exception):
-                dst.recv_checkpoint(
-                    src_rank=0,
-                    metadata=src.metadata(),
-                    step=3,
-                    timeout=timedelta(seconds=5),
-                )
+            got = dst.recv_checkpoint(
+                src_rank=0, metadata=src.metadata(), step=3,
+                timeout=timedelta(seconds=5),
+            )
+            torch.testing.assert_close(got["x"], torch.ones(4))  # snapshot value
         finally:
             src.shutdown()
             dst.shutdown()

@@ -1,0 +1,222 @@
+"""LocalSGD / DiLoCo integration tests (threads-as-replicas over gloo).
+
+Reference strategy: torchft/local_sgd_integ_test.py — healthy runs and
+recovery, asserting the global (outer) state matches across replicas.
+"""
+
+import logging
+import threading
+from concurrent.futures import ThreadPoolExecutor
+from datetime import timedelta
+from typing import Dict, List
+
+import pytest
+import torch
+import torch.nn as nn
+from torch.distributed import TCPStore
+
+from torchft_amd._ftcore import LighthouseServer
+from torchft_amd.local_sgd import DiLoCo, LocalSGD
+from torchft_amd.manager import Manager
+from torchft_amd.process_group import ProcessGroupGloo
+
+logging.getLogger("torchft_amd").setLevel(logging.WARNING)
+
+
+def _make_model() -> nn.Module:
+    torch.manual_seed(7)
+    return nn.Sequential(nn.Linear(4, 8), nn.ReLU(), nn.Linear(8, 4))
+
+
+class InjectedFailure(Exception):
+    pass
+
+
+def _localsgd_replica(replica_id: int, lighthouse_addr: str, total_outer: int,
+                      sync_every: int) -> Dict[str, torch.Tensor]:
+    store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+    model = _make_model()
+    manager = Manager(
+        pg=ProcessGroupGloo(timeout=timedelta(seconds=20)),
+        load_state_dict=model.load_state_dict,
+        state_dict=model.state_dict,
+        min_replica_size=1,
+        rank=0,
+        world_size=1,
+        store_addr="127.0.0.1",
+        store_port=store.port,
+        lighthouse_addr=lighthouse_addr,
+        replica_id=f"ls_{replica_id}",
+        hostname="127.0.0.1",
+        timeout=timedelta(seconds=20),
+    )
+    try:
+        opt = torch.optim.SGD(model.parameters(), lr=0.05)
+        criterion = nn.MSELoss()
+        with LocalSGD(manager, model, opt, sync_every=sync_every):
+            while manager.current_step() < total_outer:
+                # replica-dependent data: local models diverge between syncs
+                torch.manual_seed(manager.current_step() * 10 + replica_id)
+                x = torch.randn(4, 4)
+                y = torch.randn(4, 4)
+                opt.zero_grad()
+                criterion(model(x), y).backward()
+                opt.step()
+        return {k: v.detach().clone() for k, v in model.state_dict().items()}
+    finally:
+        manager.shutdown(wait=False)
+
+
+class TestLocalSGD:
+    def test_two_replicas_converge(self):
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=100)
+        try:
+            with ThreadPoolExecutor(max_workers=2) as ex:
+                futs = [
+                    ex.submit(_localsgd_replica, i, lh.address(), 3, 2)
+                    for i in range(2)
+                ]
+                dicts = [f.result(timeout=60) for f in futs]
+            for k in dicts[0]:
+                torch.testing.assert_close(dicts[0][k], dicts[1][k])
+        finally:
+            lh.shutdown()
+
+
+def _diloco_replica(
+    replica_id: int,
+    lighthouse_addr: str,
+    total_outer: int,
+    sync_every: int,
+    fail_at_step: int = -1,
+    attempts: int = 3,
+) -> Dict[str, object]:
+    did_fail = False
+    for attempt in range(attempts):
+        store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+        model = _make_model()
+        fragments = [model[0], model[2]]
+        inner_opt = torch.optim.SGD(model.parameters(), lr=0.05)
+        # one outer optimizer per fragment, over that fragment's params only
+        # (a shared outer optimizer would also step the non-synced fragment
+        # with stale inner grads)
+        outer_opt = [
+            torch.optim.SGD(f.parameters(), lr=0.5, momentum=0.9) for f in fragments
+        ]
+        manager = Manager(
+            pg=ProcessGroupGloo(timeout=timedelta(seconds=20)),
+            load_state_dict=model.load_state_dict,
+            state_dict=model.state_dict,
+            min_replica_size=1,
+            use_async_quorum=False,
+            rank=0,
+            world_size=1,
+            store_addr="127.0.0.1",
+            store_port=store.port,
+            lighthouse_addr=lighthouse_addr,
+            replica_id=f"dl_{replica_id}",
+            hostname="127.0.0.1",
+            timeout=timedelta(seconds=20),
+        )
+        criterion = nn.MSELoss()
+        fail_armed = fail_at_step >= 0 and attempt == 0
+        try:
+            diloco = DiLoCo(
+                manager,
+                fragments,
+                inner_opt,
+                outer_opt,
+                sync_every=sync_every,
+                pin_memory=False,
+            )
+            with diloco:
+                local_batches = 0
+                while manager.current_step() < total_outer:
+                    if fail_armed and manager.current_step() >= fail_at_step:
+                        raise InjectedFailure(f"fail replica {replica_id}")
+                    torch.manual_seed(local_batches * 10 + replica_id)
+                    x = torch.randn(4, 4)
+                    y = torch.randn(4, 4)
+                    inner_opt.zero_grad()
+                    criterion(model(x), y).backward()
+                    inner_opt.step()
+                    local_batches += 1
+            return {
+                "original": {
+                    f"{i}_{k}": v.detach().clone()
+                    for i, frag in enumerate(diloco._fragments)
+                    for k, v in frag.original_parameters.items()
+                },
+                "failed": did_fail,
+            }
+        except InjectedFailure:
+            did_fail = True
+            continue
+        finally:
+            manager.shutdown(wait=False)
+    raise RuntimeError("unreachable")
+
+
+class TestDiLoCo:
+    def test_two_replicas_healthy(self):
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=100)
+        try:
+            with ThreadPoolExecutor(max_workers=2) as ex:
+                futs = [
+                    ex.submit(_diloco_replica, i, lh.address(), 2, 4)
+                    for i in range(2)
+                ]
+                results = [f.result(timeout=90) for f in futs]
+            a, b = results[0]["original"], results[1]["original"]
+            for k in a:
+                torch.testing.assert_close(a[k], b[k], msg=f"mismatch at {k}")
+        finally:
+            lh.shutdown()
+
+    def test_validation_rejects_async_quorum(self):
+        store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=50)
+        model = _make_model()
+        manager = Manager(
+            pg=ProcessGroupGloo(),
+            load_state_dict=model.load_state_dict,
+            state_dict=model.state_dict,
+            min_replica_size=1,
+            use_async_quorum=True,
+            rank=0,
+            world_size=1,
+            store_addr="127.0.0.1",
+            store_port=store.port,
+            lighthouse_addr=lh.address(),
+            replica_id="v0",
+            hostname="127.0.0.1",
+        )
+        try:
+            with pytest.raises(ValueError, match="synchronous quorum"):
+                DiLoCo(
+                    manager,
+                    [model],
+                    torch.optim.SGD(model.parameters(), lr=0.1),
+                    torch.optim.SGD(model.parameters(), lr=0.1),
+                    sync_every=2,
+                    pin_memory=False,
+                )
+        finally:
+            manager.shutdown(wait=False)
+            lh.shutdown()
+
+    def test_recovery_after_failure(self):
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=100)
+        try:
+            with ThreadPoolExecutor(max_workers=2) as ex:
+                futs = [
+                    ex.submit(_diloco_replica, 0, lh.address(), 3, 2),
+                    ex.submit(_diloco_replica, 1, lh.address(), 3, 2, 1),
+                ]
+                results = [f.result(timeout=120) for f in futs]
+            assert results[1]["failed"]
+            a, b = results[0]["original"], results[1]["original"]
+            for k in a:
+                torch.testing.assert_close(a[k], b[k], msg=f"mismatch at {k}")
+        finally:
+            lh.shutdown()

@@ -168,22 +168,23 @@ class HTTPTransport(CheckpointTransport[T], Generic[T]):
         return f"http://{_local_hostname()}:{self._port}"
 
     def _stage(self, state_dict: T) -> object:
-        """Copy accelerator tensors to CPU on the staging HIP stream."""
+        """Snapshot the state dict: device tensors copy to pinned CPU on the
+        staging HIP stream; CPU tensors clone. The snapshot must not alias
+        live training state — the owner keeps mutating it while recovering
+        replicas fetch."""
         leaves, spec = tree_flatten(state_dict)
-        if not torch.cuda.is_available():
-            return state_dict
         out = []
         with get_stream_context(self._staging_stream):
             for leaf in leaves:
-                if isinstance(leaf, torch.Tensor) and leaf.device.type != "cpu":
-                    cpu = torch.empty_strided(
-                        leaf.shape,
-                        leaf.stride() if leaf.is_contiguous() else leaf.contiguous().stride(),
-                        dtype=leaf.dtype,
-                        pin_memory=True,
-                    )
-                    cpu.copy_(leaf.detach().contiguous(), non_blocking=True)
-                    out.append(cpu)
+                if isinstance(leaf, torch.Tensor):
+                    if leaf.device.type != "cpu":
+                        cpu = torch.empty(
+                            leaf.shape, dtype=leaf.dtype, pin_memory=True
+                        )
+                        cpu.copy_(leaf.detach().contiguous(), non_blocking=True)
+                        out.append(cpu)
+                    else:
+                        out.append(leaf.detach().clone())
                 else:
                     out.append(leaf)
         if self._staging_stream is not None:
@@ -199,8 +200,17 @@ class HTTPTransport(CheckpointTransport[T], Generic[T]):
             self._allowed_step = step
 
     def disallow_checkpoint(self) -> None:
+        """No-op by design: the staged checkpoint is a full snapshot (no
+        aliasing of live training state), so serving it after training
+        resumes is safe and removes the fetch-vs-next-step race the
+        reference's live-serving transport has. The snapshot is replaced at
+        the next send_checkpoint."""
+        pass
+
+    def _drop_checkpoint(self) -> None:
         with self._lock.w_lock():
             self._allowed_step = None
+            self._staged = None
 
     def recv_checkpoint(
         self, src_rank: int, metadata: str, step: int, timeout: timedelta

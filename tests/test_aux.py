@@ -1,0 +1,95 @@
+"""Aux runtime tests: telemetry, parameter server, launcher component,
+futures timeouts."""
+
+import json
+import logging
+import time
+from datetime import timedelta
+
+import pytest
+import torch
+
+from torchft_amd.futures import context_timeout, future_timeout
+from torchft_amd.telemetry import JSONLineFormatter, setup_telemetry
+
+
+class TestFutures:
+    def test_future_timeout_fires(self):
+        fut: torch.futures.Future = torch.futures.Future()
+        timed = future_timeout(fut, timedelta(milliseconds=100))
+        with pytest.raises(TimeoutError):
+            timed.wait()
+
+    def test_future_timeout_passthrough(self):
+        fut: torch.futures.Future = torch.futures.Future()
+        timed = future_timeout(fut, timedelta(seconds=5))
+        fut.set_result(42)
+        assert timed.wait() == 42
+
+    def test_future_timeout_propagates_exception(self):
+        fut: torch.futures.Future = torch.futures.Future()
+        timed = future_timeout(fut, timedelta(seconds=5))
+        fut.set_exception(ValueError("boom"))
+        with pytest.raises(ValueError, match="boom"):
+            timed.wait()
+
+    def test_context_timeout_fires(self):
+        fired = []
+        with context_timeout(lambda: fired.append(1), timedelta(milliseconds=50)):
+            time.sleep(0.3)
+        assert fired == [1]
+
+    def test_context_timeout_cancelled(self):
+        fired = []
+        with context_timeout(lambda: fired.append(1), timedelta(seconds=5)):
+            pass
+        time.sleep(0.1)
+        assert fired == []
+
+
+class TestTelemetry:
+    def test_json_formatter_includes_extras(self):
+        fmt = JSONLineFormatter()
+        record = logging.LogRecord(
+            "torchft_quorums", logging.INFO, __file__, 1, "", (), None
+        )
+        record.quorum_id = 7
+        record.replica_id = "rep0"
+        out = json.loads(fmt.format(record))
+        assert out["quorum_id"] == 7
+        assert out["replica_id"] == "rep0"
+        assert out["logger"] == "torchft_quorums"
+
+
+class TestParameterServer:
+    def test_session_roundtrip(self):
+        from torchft_amd.parameter_server import ParameterServer
+        from torchft_amd.process_group import ProcessGroup
+
+        class EchoPS(ParameterServer):
+            def forward(self, session_id: str, pg: ProcessGroup) -> None:
+                t = torch.arange(8.0)
+                pg.send([t], 1, tag=0).wait()
+
+        ps = EchoPS()
+        try:
+            pg = EchoPS.connect(ps.address(), timeout=timedelta(seconds=10))
+            t = torch.zeros(8)
+            pg.recv([t], 0, tag=0).wait()
+            torch.testing.assert_close(t, torch.arange(8.0))
+        finally:
+            ps.shutdown()
+
+
+class TestLauncherComponent:
+    def test_build_replica_cmd(self):
+        from torchft_amd.launcher import build_replica_cmd
+
+        cmd, env = build_replica_cmd(
+            ["train.py", "--steps", "10"], 1, 4, 2, "http://lh:123", 29650
+        )
+        assert "--nproc-per-node=2" in cmd
+        assert cmd[-3:] == ["train.py", "--steps", "10"]
+        assert env["REPLICA_GROUP_ID"] == "1"
+        assert env["NUM_REPLICA_GROUPS"] == "4"
+        assert env["TORCHFT_LIGHTHOUSE"] == "http://lh:123"

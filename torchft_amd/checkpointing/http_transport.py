@@ -212,20 +212,31 @@ class HTTPTransport(CheckpointTransport[T], Generic[T]):
             self._allowed_step = None
             self._staged = None
 
+    def _open_with_retry(self, url: str, timeout: timedelta):
+        """GET with retry on 400: the source stages its snapshot in its own
+        quorum thread, which may lag the destination's fetch by a moment."""
+        import time as _time
+        import urllib.error
+
+        deadline = _time.monotonic() + timeout.total_seconds()
+        while True:
+            try:
+                return urllib.request.urlopen(url, timeout=timeout.total_seconds())
+            except urllib.error.HTTPError as e:
+                if e.code != 400 or _time.monotonic() > deadline - 0.2:
+                    raise
+                _time.sleep(0.05)
+
     def recv_checkpoint(
         self, src_rank: int, metadata: str, step: int, timeout: timedelta
     ) -> T:
         base = f"{metadata}/checkpoint/{step}"
         if self._num_chunks <= 0:
-            with urllib.request.urlopen(
-                f"{base}/full", timeout=timeout.total_seconds()
-            ) as resp:
+            with self._open_with_retry(f"{base}/full", timeout) as resp:
                 return streaming_load(resp)
 
         # chunked parallel fetch: header first, then N ranges concurrently
-        with urllib.request.urlopen(
-            f"{base}/metadata", timeout=timeout.total_seconds()
-        ) as resp:
+        with self._open_with_retry(f"{base}/metadata", timeout) as resp:
             spec, metas = pickle.loads(resp.read())
 
         tensor_metas = [m for m in metas if not isinstance(m, _PickledLeaf)]

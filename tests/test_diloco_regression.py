@@ -1,0 +1,129 @@
+"""DiLoCo numerical regression harness with golden fixtures.
+
+Reference strategy: torchft/diloco_regression_test.py — deterministic
+MockModel/updates, per-outer-step parameter trajectories compared against
+golden JSON in test_fixtures/; regenerate with WRITE_FIXTURE=true.
+"""
+
+import json
+import os
+from concurrent.futures import ThreadPoolExecutor
+from datetime import timedelta
+from typing import Dict, List
+
+import torch
+import torch.nn as nn
+from torch.distributed import TCPStore
+
+from torchft_amd._ftcore import LighthouseServer
+from torchft_amd.local_sgd import DiLoCo
+from torchft_amd.manager import Manager
+from torchft_amd.process_group import ProcessGroupGloo
+
+FIXTURE_PATH = os.path.join(
+    os.path.dirname(os.path.abspath(__file__)), "test_fixtures", "diloco_trajectory.json"
+)
+WRITE_FIXTURE = os.environ.get("WRITE_FIXTURE", "false").lower() == "true"
+
+
+def _mock_model() -> nn.Module:
+    # deterministic weights WITHOUT the global RNG (replicas run as threads
+    # in one process; manual_seed would race between them)
+    model = nn.Sequential(nn.Linear(3, 4, bias=False), nn.Linear(4, 2, bias=False))
+    with torch.no_grad():
+        for i, p in enumerate(model.parameters()):
+            vals = torch.arange(p.numel(), dtype=torch.float32).reshape(p.shape)
+            p.copy_(vals * 0.01 - 0.03 * (i + 1))
+    return model
+
+
+def _deterministic_grad(step: int, replica: int, p: torch.Tensor) -> torch.Tensor:
+    """Deterministic pseudo-data gradient: no RNG races between threads."""
+    g = torch.arange(p.numel(), dtype=torch.float32).reshape(p.shape)
+    return (g + step) * 0.01 * (replica + 1)
+
+
+def _run_replica(replica_id: int, lighthouse_addr: str, outer_steps: int) -> List[Dict]:
+    store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+    model = _mock_model()
+    fragments = [model[0], model[1]]
+    inner_opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    outer_opts = [
+        torch.optim.SGD(f.parameters(), lr=0.5, momentum=0.9) for f in fragments
+    ]
+    manager = Manager(
+        pg=ProcessGroupGloo(timeout=timedelta(seconds=20)),
+        load_state_dict=model.load_state_dict,
+        state_dict=model.state_dict,
+        min_replica_size=2,
+        use_async_quorum=False,
+        init_sync=False,  # identical seeds; keep trajectories pure
+        rank=0,
+        world_size=1,
+        store_addr="127.0.0.1",
+        store_port=store.port,
+        lighthouse_addr=lighthouse_addr,
+        replica_id=f"fix_{replica_id}",
+        hostname="127.0.0.1",
+        timeout=timedelta(seconds=20),
+    )
+    trajectory: List[Dict] = []
+    try:
+        diloco = DiLoCo(
+            manager, fragments, inner_opt, outer_opts, sync_every=2, pin_memory=False
+        )
+        with diloco:
+            step = 0
+            while manager.current_step() < outer_steps:
+                for p in model.parameters():
+                    p.grad = _deterministic_grad(step, replica_id, p)
+                inner_opt.step()
+                step += 1
+                if step % 2 == 0:  # just synced: record the global params
+                    trajectory.append(
+                        {
+                            f"{i}_{name}": param.tolist()
+                            for i, frag in enumerate(diloco._fragments)
+                            for name, param in frag.original_parameters.items()
+                        }
+                    )
+        return trajectory
+    finally:
+        manager.shutdown(wait=False)
+
+
+class TestDiLoCoRegression:
+    def test_trajectory_matches_fixture(self):
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=100)
+        try:
+            with ThreadPoolExecutor(max_workers=2) as ex:
+                futs = [ex.submit(_run_replica, i, lh.address(), 3) for i in range(2)]
+                trajs = [f.result(timeout=90) for f in futs]
+        finally:
+            lh.shutdown()
+
+        # both replicas must follow the identical global trajectory
+        assert trajs[0] == trajs[1], "replica trajectories diverged"
+
+        if WRITE_FIXTURE:
+            os.makedirs(os.path.dirname(FIXTURE_PATH), exist_ok=True)
+            with open(FIXTURE_PATH, "w") as f:
+                json.dump(trajs[0], f, indent=1)
+            return
+
+        assert os.path.exists(FIXTURE_PATH), (
+            f"fixture missing; regenerate with WRITE_FIXTURE=true ({FIXTURE_PATH})"
+        )
+        with open(FIXTURE_PATH) as f:
+            golden = json.load(f)
+        assert len(golden) == len(trajs[0])
+        for got_step, golden_step in zip(trajs[0], golden):
+            assert set(got_step) == set(golden_step)
+            for k in golden_step:
+                torch.testing.assert_close(
+                    torch.tensor(got_step[k]),
+                    torch.tensor(golden_step[k]),
+                    rtol=1e-5,
+                    atol=1e-6,
+                    msg=f"regression at {k}",
+                )

@@ -1,0 +1,271 @@
+"""Manager unit tests with a mocked coordination client.
+
+Reference strategy: torchft/manager_test.py — drive quorum/heal/commit
+paths with a MagicMock ManagerClient and ProcessGroupDummy, no servers.
+"""
+
+from datetime import timedelta
+from typing import Optional
+from unittest.mock import MagicMock, patch
+
+import pytest
+import torch
+from torch.distributed import TCPStore
+
+from torchft_amd._ftcore import QuorumResult
+from torchft_amd.manager import Manager, WorldSizeMode
+from torchft_amd.process_group import ProcessGroupDummy
+
+
+def mock_quorum(
+    quorum_id=1,
+    replica_rank=0,
+    replica_world_size=2,
+    recover_src_manager_address="",
+    recover_src_replica_rank: Optional[int] = None,
+    recover_dst_replica_ranks=(),
+    store_address="127.0.0.1:0",
+    max_step=0,
+    max_replica_rank: Optional[int] = 0,
+    max_world_size=2,
+    heal=False,
+    commit_failures=0,
+    replica_ids=("a", "b"),
+) -> QuorumResult:
+    q = QuorumResult()
+    q.quorum_id = quorum_id
+    q.replica_rank = replica_rank
+    q.replica_world_size = replica_world_size
+    q.recover_src_manager_address = recover_src_manager_address
+    q.recover_src_replica_rank = recover_src_replica_rank
+    q.recover_dst_replica_ranks = list(recover_dst_replica_ranks)
+    q.store_address = store_address
+    q.max_step = max_step
+    q.max_replica_rank = max_replica_rank
+    q.max_world_size = max_world_size
+    q.heal = heal
+    q.commit_failures = commit_failures
+    q.replica_ids = list(replica_ids)
+    return q
+
+
+def make_manager(client: MagicMock, use_async_quorum=True, **kwargs) -> Manager:
+    store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+    state = {"w": torch.zeros(4)}
+    with patch("torchft_amd.manager.ManagerClient", return_value=client), patch(
+        "torchft_amd.manager.ManagerServer"
+    ) as mock_server:
+        mock_server.return_value.address.return_value = "http://127.0.0.1:1"
+        manager = Manager(
+            pg=ProcessGroupDummy(0, 1),
+            load_state_dict=lambda sd: state.update(sd),
+            state_dict=lambda: state,
+            min_replica_size=1,
+            use_async_quorum=use_async_quorum,
+            rank=0,
+            world_size=1,
+            store_addr="127.0.0.1",
+            store_port=store.port,
+            lighthouse_addr="http://nowhere:1",
+            replica_id="test1",
+            hostname="127.0.0.1",
+            timeout=timedelta(seconds=5),
+            **kwargs,
+        )
+    manager._test_store = store  # keep the store alive
+    manager._test_state = state
+    return manager
+
+
+class TestManagerQuorum:
+    def test_healthy_step_commits(self):
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum()
+        client.should_commit.return_value = True
+        m = make_manager(client)
+        try:
+            m.start_quorum()
+            t = torch.ones(4)
+            work = m.allreduce(t)
+            assert work.wait()
+            # AVG normalization by participants (max_world_size=2)
+            torch.testing.assert_close(t, torch.full((4,), 0.5))
+            assert m.should_commit()
+            assert m.current_step() == 1
+            assert m.batches_committed() == 2
+        finally:
+            m.shutdown(wait=False)
+
+    def test_quorum_id_change_reconfigures_pg(self):
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum(quorum_id=5)
+        client.should_commit.return_value = True
+        m = make_manager(client)
+        try:
+            m.start_quorum()
+            m.wait_quorum()
+            assert m._pg.configure_count == 1
+            # same quorum id -> no reconfigure
+            m.should_commit()
+            m.start_quorum()
+            m.wait_quorum()
+            assert m._pg.configure_count == 1
+            # new quorum id -> reconfigure
+            client._quorum.return_value = mock_quorum(quorum_id=6, max_step=1)
+            m.should_commit()
+            m.start_quorum()
+            m.wait_quorum()
+            assert m._pg.configure_count == 2
+        finally:
+            m.shutdown(wait=False)
+
+    def test_error_zeros_and_rejects_commit(self):
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum()
+        client.should_commit.return_value = False
+        m = make_manager(client)
+        try:
+            m.start_quorum()
+            m.wait_quorum()
+            m.report_error(RuntimeError("boom"))
+            t = torch.ones(4)
+            work = m.allreduce(t)  # becomes a no-op after error
+            assert work.wait()
+            assert not m.should_commit()
+            assert m.current_step() == 0
+            # client is told should_commit=False
+            assert client.should_commit.call_args[0][2] is False
+        finally:
+            m.shutdown(wait=False)
+
+    def test_healing_async_not_participating(self):
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum(
+            replica_rank=1,
+            recover_src_manager_address="http://src:1",
+            recover_src_replica_rank=0,
+            max_step=5,
+            max_replica_rank=None,
+            max_world_size=1,
+            heal=True,
+        )
+        client.should_commit.return_value = True
+        src_client = MagicMock()
+        src_client._checkpoint_metadata.return_value = "http://src-ckpt:1"
+
+        m = make_manager(client)
+        try:
+            transport = MagicMock()
+            transport.metadata.return_value = "http://me:1"
+            transport.recv_checkpoint.return_value = {
+                "user": {"default": {"w": torch.full((4,), 7.0)}},
+                "torchft": {"step": 5, "batches_committed": 10},
+            }
+            m._checkpoint_transport = transport
+            with patch("torchft_amd.manager.ManagerClient", return_value=src_client):
+                m.start_quorum()
+                m.wait_quorum()
+            assert m._healing
+            assert not m.is_participating()
+            assert m.num_participants() == 1
+            t = torch.ones(4)
+            m.allreduce(t).wait()
+            # non-participating grads are zeroed
+            torch.testing.assert_close(t, torch.zeros(4))
+            assert m.should_commit()
+            # healed state applied on the main thread at should_commit
+            torch.testing.assert_close(m._test_state["w"], torch.full((4,), 7.0))
+            assert m.current_step() == 6  # healed to 5, committed one step
+        finally:
+            m.shutdown(wait=False)
+
+    def test_sync_quorum_applies_state_eagerly(self):
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum(
+            replica_rank=1,
+            recover_src_manager_address="http://src:1",
+            recover_src_replica_rank=0,
+            max_step=3,
+            max_replica_rank=None,
+            heal=True,
+        )
+        src_client = MagicMock()
+        src_client._checkpoint_metadata.return_value = "meta"
+        m = make_manager(client, use_async_quorum=False)
+        try:
+            transport = MagicMock()
+            transport.metadata.return_value = "m"
+            transport.recv_checkpoint.return_value = {
+                "user": {"default": {"w": torch.full((4,), 3.0)}},
+                "torchft": {"step": 3, "batches_committed": 6},
+            }
+            m._checkpoint_transport = transport
+            with patch("torchft_amd.manager.ManagerClient", return_value=src_client):
+                m.start_quorum()
+            # state applied eagerly; participating in sync mode
+            torch.testing.assert_close(m._test_state["w"], torch.full((4,), 3.0))
+            assert not m._healing
+            assert m.is_participating()
+            assert m.current_step() == 3
+        finally:
+            m.shutdown(wait=False)
+
+    def test_fixed_with_spares_caps_world(self):
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum(
+            replica_rank=1, max_replica_rank=1, max_world_size=2, replica_world_size=2
+        )
+        m = make_manager(
+            client,
+            world_size_mode=WorldSizeMode.FIXED_WITH_SPARES,
+        )
+        # min_replica_size=1 -> only 1 participates; rank 1 is a spare
+        try:
+            m.start_quorum()
+            m.wait_quorum()
+            assert m.num_participants() == 1
+            assert m.participating_rank() is None
+        finally:
+            m.shutdown(wait=False)
+
+    def test_max_retries_raises(self):
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum()
+        client.should_commit.return_value = False
+        m = make_manager(client, max_retries=1)
+        try:
+            m.start_quorum()
+            assert not m.should_commit()  # failure 1
+            m.start_quorum()
+            with pytest.raises(RuntimeError, match="max_retries"):
+                m.should_commit()  # failure 2 > max_retries
+        finally:
+            m.shutdown(wait=False)
+
+    def test_commit_failures_forwarded_to_quorum(self):
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum()
+        client.should_commit.return_value = False
+        m = make_manager(client)
+        try:
+            m.start_quorum()
+            m.should_commit()
+            m.start_quorum()
+            m.wait_quorum()
+            assert client._quorum.call_args.kwargs["commit_failures"] == 1
+        finally:
+            m.shutdown(wait=False)
+
+    def test_report_error_and_wrap_future(self):
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum()
+        m = make_manager(client)
+        try:
+            fut: torch.futures.Future = torch.futures.Future()
+            wrapped = m.wrap_future(fut, default=torch.zeros(2))
+            fut.set_exception(RuntimeError("inner fail"))
+            out = wrapped.wait()
+            torch.testing.assert_close(out, torch.zeros(2))
+            assert m.errored() is not None
+        finally:
+            m.shutdown(wait=False)

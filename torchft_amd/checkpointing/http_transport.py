@@ -36,6 +36,7 @@ from torchft_amd.checkpointing._serialization import (
     _readinto_tensor,
     _read_exact,
     _tensor_bytes,
+    save_plan,
     split_state_dict,
     streaming_load,
     streaming_save,
@@ -123,14 +124,18 @@ class HTTPTransport(CheckpointTransport[T], Generic[T]):
                 import io
 
                 if what == "full":
-                    buf = io.BytesIO()
-                    streaming_save(obj, buf)
-                    data = buf.getvalue()
+                    # stream tensors straight into the socket (zero copy,
+                    # no whole-body buffer) with an exact content length
+                    prefix, tensors, total = save_plan(obj)
                     self.send_response(200)
                     self.send_header("Content-Type", "application/octet-stream")
-                    self.send_header("Content-Length", str(len(data)))
+                    self.send_header("Content-Length", str(total))
                     self.end_headers()
-                    self.wfile.write(data)
+                    self.wfile.write(prefix)
+                    for t in tensors:
+                        raw = _tensor_bytes(t)
+                        if raw.numel():
+                            self.wfile.write(memoryview(raw.numpy()))
                     return
                 metas, tensors, spec = split_state_dict(obj)
                 if what == "metadata":

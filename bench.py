@@ -180,6 +180,7 @@ def main() -> None:
             connect_timeout=timedelta(seconds=60),
             # identical random init on every replica; no step-0 state transfer
             init_sync=False,
+            should_quantize=args.quantize,
         )
 
     dbg("manager ready")

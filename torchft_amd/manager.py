@@ -127,6 +127,7 @@ class Manager:
         init_sync: bool = True,
         max_retries: Optional[int] = None,
         quorum_retries: int = 0,
+        should_quantize: bool = False,
     ) -> None:
         self.quorum_logger: logging.Logger = logging.getLogger("torchft_quorums")
         self.commits_logger: logging.Logger = logging.getLogger("torchft_commits")
@@ -154,6 +155,7 @@ class Manager:
             os.environ.get(CONNECT_TIMEOUT_SEC_ENV), connect_timeout
         )
 
+        self._default_should_quantize = should_quantize
         self._replica_world_size_mode = world_size_mode
         self._init_sync = init_sync
         self._max_retries = max_retries
@@ -301,7 +303,7 @@ class Manager:
     def allreduce(
         self,
         tensor: torch.Tensor,
-        should_quantize: bool = False,
+        should_quantize: Optional[bool] = None,
         reduce_op: ReduceOp = ReduceOp.AVG,
     ) -> Work:
         """Fault-tolerant allreduce; AVG scales by 1/num_participants.
@@ -313,6 +315,9 @@ class Manager:
         ``should_quantize=True`` routes through the fp8 quantized allreduce
         (CDNA4 HIP kernels + alltoall/allgather over all 7 xGMI links).
         """
+        if should_quantize is None:
+            should_quantize = self._default_should_quantize
+
         if self.errored():
             return _DummyWork(tensor)
 

@@ -45,6 +45,10 @@ def parse_args() -> argparse.Namespace:
                    help="disable the fault-tolerance control plane (raw perf)")
     p.add_argument("--quantize", action="store_true",
                    help="fp8-quantized cross-replica allreduce")
+    p.add_argument("--checkpoint-activations", choices=["auto", "on", "off"],
+                   default="auto",
+                   help="activation checkpointing; auto = off for <=8B on "
+                        "MI355X (288 GB HBM fits the full activations), on for 70B")
     return p.parse_args()
 
 
@@ -139,7 +143,13 @@ def main() -> None:
     seq = min(args.seq, cfg.max_seq_len)
     torch.manual_seed(1234)
     dtype = torch.bfloat16
-    model = Llama(cfg, dtype=dtype, checkpoint_activations=args.model != "debug")
+    if args.checkpoint_activations == "auto":
+        # 288 GB HBM3E: 8B-class activations (~2 GB/layer at bs2 x 8k) fit
+        # without recompute; 70B needs checkpointing
+        ckpt = args.model == "llama3_70b"
+    else:
+        ckpt = args.checkpoint_activations == "on"
+    model = Llama(cfg, dtype=dtype, checkpoint_activations=ckpt)
     model = model.to(device)
 
     # ---- fault-tolerance control plane ------------------------------------

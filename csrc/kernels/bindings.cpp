@@ -239,6 +239,45 @@ void adamw_step(const std::vector<at::Tensor>& params,
                     (float)weight_decay, bc1, bc2, (tft_stream)stream);
 }
 
+// ---- flash attention backward (appended) -----------------------------------
+
+std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                               at::Tensor dout, at::Tensor lse, at::Tensor delta,
+                               double scale, bool causal) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 && q.dim() == 4);
+  TORCH_CHECK(q.size(3) == 128, "fa_bwd supports head_dim=128 only");
+  TORCH_CHECK(q.size(2) % 128 == 0, "fa_bwd requires seq % 128 == 0");
+  const int64_t B = q.size(0), Hq = q.size(1), S = q.size(2);
+  const int64_t Hkv = k.size(1);
+  TORCH_CHECK(Hq % Hkv == 0, "Hq must be a multiple of Hkv");
+  auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous(),
+       doc = dout.contiguous();
+  auto lsec = lse.contiguous().to(at::kFloat);
+  auto deltac = delta.contiguous().to(at::kFloat);
+  TORCH_CHECK(lsec.numel() == B * Hq * S, "lse shape mismatch");
+  auto dq = at::empty_like(qc);
+  auto dk = at::empty_like(kc);
+  auto dv = at::empty_like(vc);
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  tft::launch_fa_bwd(qc.data_ptr(), kc.data_ptr(), vc.data_ptr(), doc.data_ptr(),
+                     lsec.data_ptr<float>(), deltac.data_ptr<float>(),
+                     dq.data_ptr(), dk.data_ptr(), dv.data_ptr(), (int)B,
+                     (int)Hq, (int)Hkv, (int)S, (float)scale, causal,
+                     (tft_stream)stream);
+  return {dq, dk, dv};
+}
+
+at::Tensor mfma_probe(at::Tensor A, at::Tensor B) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(A.sizes() == at::IntArrayRef({32, 16}) &&
+              B.sizes() == at::IntArrayRef({16, 32}));
+  auto D = at::zeros({32, 32}, A.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  tft::launch_mfma_probe(A.contiguous().data_ptr(), B.contiguous().data_ptr(),
+                         D.data_ptr<float>(), (tft_stream)stream);
+  return D;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "torchft_amd CDNA4 (gfx950) HIP kernels";
   m.def("fp8_pack_bytes", &fp8_pack_bytes);
@@ -252,4 +291,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("adamw_step", &adamw_step);
+  m.def("fa_bwd", &fa_bwd,
+        "flash-attention backward (bf16, D=128): returns (dq, dk, dv)");
+  m.def("mfma_probe", &mfma_probe, "mfma_f32_32x32x16_bf16 layout probe");
 }
+

@@ -53,3 +53,16 @@ void launch_adamw(const int64_t* param_ptrs, const int64_t* grad_ptrs,
                   tft_stream stream);
 
 }  // namespace torchft_amd
+
+namespace torchft_amd {
+
+// flash_attn_bwd.hip --------------------------------------------------------
+void launch_fa_bwd(const void* q, const void* k, const void* v, const void* dout,
+                   const float* lse, const float* delta, void* dq, void* dk,
+                   void* dv, int B, int Hq, int Hkv, int S, float scale,
+                   bool causal, tft_stream stream);
+
+// mfma_probe.hip ------------------------------------------------------------
+void launch_mfma_probe(const void* A, const void* B, float* D, tft_stream s);
+
+}  // namespace torchft_amd

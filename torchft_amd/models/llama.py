@@ -18,6 +18,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from torchft_amd.ops import RMSNorm, rope, rope_tables, swiglu
+from torchft_amd.ops.flash_attention import flash_attention
 
 
 @dataclass
@@ -65,9 +66,9 @@ class Attention(nn.Module):
         v = self.wv(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
         q = rope(q, cos, sin)
         k = rope(k, cos, sin)
-        # SDPA wants [B, H, S, D]
-        q, k, v = (t.transpose(1, 2) for t in (q, k, v))
-        out = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)
+        # SDPA layout [B, H, S, D]; custom CDNA4 backward where supported
+        q, k, v = (t.transpose(1, 2).contiguous() for t in (q, k, v))
+        out = flash_attention(q, k, v, causal=True)
         out = out.transpose(1, 2).reshape(B, S, -1)
         return self.wo(out)
 

@@ -1,0 +1,82 @@
+"""Flash attention with a hand-written CDNA4 backward.
+
+Forward: aten's flash kernel (fast on gfx950, and it hands us the
+logsumexp). Backward: the MFMA kernels in csrc/kernels/flash_attn_bwd.hip —
+the stock backward is the single largest kernel of the Llama-8B step
+(profiles/SUMMARY.md).
+
+Constraints of the custom backward: bf16, head_dim=128, seq % 128 == 0,
+dropout 0. Anything else falls back to stock SDPA. Enable/disable with
+TORCHFT_AMD_CUSTOM_FA=1/0 (default on when the extension is present).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from torchft_amd.ops import hip_ext
+
+_ENV = "TORCHFT_AMD_CUSTOM_FA"
+
+
+def custom_fa_enabled() -> bool:
+    v = os.environ.get(_ENV, "1")
+    return v not in ("0", "false", "False")
+
+
+class _FlashAttentionFn(torch.autograd.Function):
+    """q,k,v: [B, H, S, D] (SDPA layout), causal, GQA-native."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal: bool, scale: float):
+        out, lse, *_ = torch.ops.aten._scaled_dot_product_flash_attention(
+            q, k, v, 0.0, causal, False, scale=scale
+        )
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.causal = causal
+        ctx.scale = scale
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse = ctx.saved_tensors
+        delta = (dout.float() * out.float()).sum(-1)  # [B, Hq, S]
+        dq, dk, dv = hip_ext().fa_bwd(
+            q, k, v, dout, lse, delta, ctx.scale, ctx.causal
+        )
+        return dq, dk, dv, None, None
+
+
+def _supported(q: torch.Tensor, k: torch.Tensor) -> bool:
+    return (
+        q.is_cuda
+        and q.dtype == torch.bfloat16
+        and q.shape[-1] == 128
+        and q.shape[2] % 128 == 0
+        and k.shape[2] == q.shape[2]
+        and q.shape[1] % k.shape[1] == 0
+        and custom_fa_enabled()
+        and hip_ext() is not None
+    )
+
+
+def flash_attention(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    causal: bool = True,
+    scale: Optional[float] = None,
+) -> torch.Tensor:
+    """SDPA-compatible attention ([B, H, S, D]) with the custom backward
+    where supported, stock SDPA otherwise."""
+    if scale is None:
+        scale = q.shape[-1] ** -0.5
+    if _supported(q, k):
+        return _FlashAttentionFn.apply(q, k, v, causal, scale)
+    return F.scaled_dot_product_attention(
+        q, k, v, is_causal=causal, scale=scale, enable_gqa=q.shape[1] != k.shape[1]
+    )

@@ -32,7 +32,9 @@ def run(custom: bool, iters=10):
     return (time.perf_counter() - t0) / iters * 1000
 
 # interleaved A/B (guide rule 24)
-for rnd in range(3):
+import os
+ROUNDS = int(os.environ.get("FA_ROUNDS", "3"))
+for rnd in range(ROUNDS):
     a = run(False); b = run(True)
     fl_fwd = 2*2*B*Hq*S*S*D/2
     fl_bwd = fl_fwd * 2.5

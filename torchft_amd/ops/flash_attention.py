@@ -6,8 +6,15 @@ the stock backward is the single largest kernel of the Llama-8B step
 (profiles/SUMMARY.md).
 
 Constraints of the custom backward: bf16, head_dim=128, seq % 128 == 0,
-dropout 0. Anything else falls back to stock SDPA. Enable/disable with
-TORCHFT_AMD_CUSTOM_FA=1/0 (default on when the extension is present).
+dropout 0. Anything else falls back to stock SDPA.
+
+Status (round 1, measured on MI355X at the 8B bench shape): numerically
+correct (tests/test_flash_attn_gpu.py) but 1.34x slower than the stock
+backward — 17.6 ms vs 13.2 ms fwd+bwd. The LDS-conflict swizzle fix took
+it from 21.2 to 17.6 ms (bank conflicts were ~6 cycles per LDS
+instruction); the identified next levers are ds_read_b64_tr_b16 instead of
+the transposed LDS images and larger KV tiles to amortize staging. Until
+it wins, it is OPT-IN: set TORCHFT_AMD_CUSTOM_FA=1 (default off).
 """
 
 from __future__ import annotations
@@ -24,8 +31,8 @@ _ENV = "TORCHFT_AMD_CUSTOM_FA"
 
 
 def custom_fa_enabled() -> bool:
-    v = os.environ.get(_ENV, "1")
-    return v not in ("0", "false", "False")
+    v = os.environ.get(_ENV, "0")
+    return v in ("1", "true", "True")
 
 
 class _FlashAttentionFn(torch.autograd.Function):

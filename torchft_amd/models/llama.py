@@ -66,8 +66,9 @@ class Attention(nn.Module):
         v = self.wv(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
         q = rope(q, cos, sin)
         k = rope(k, cos, sin)
-        # SDPA layout [B, H, S, D]; custom CDNA4 backward where supported
-        q, k, v = (t.transpose(1, 2).contiguous() for t in (q, k, v))
+        # SDPA layout [B, H, S, D]; custom CDNA4 backward where enabled
+        # (stock SDPA handles the transposed views without a copy)
+        q, k, v = (t.transpose(1, 2) for t in (q, k, v))
         out = flash_attention(q, k, v, causal=True)
         out = out.transpose(1, 2).reshape(B, S, -1)
         return self.wo(out)

@@ -83,7 +83,9 @@ def flash_attention(
     if scale is None:
         scale = q.shape[-1] ** -0.5
     if _supported(q, k):
-        return _FlashAttentionFn.apply(q, k, v, causal, scale)
+        return _FlashAttentionFn.apply(
+            q.contiguous(), k.contiguous(), v.contiguous(), causal, scale
+        )
     return F.scaled_dot_product_attention(
         q, k, v, is_causal=causal, scale=scale, enable_gqa=q.shape[1] != k.shape[1]
     )

@@ -39,7 +39,7 @@ def parse_args() -> argparse.Namespace:
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", default="llama3_8b",
                    choices=["llama3_8b", "llama3_70b", "debug"])
-    p.add_argument("--batch", type=int, default=2, help="per-GPU batch size")
+    p.add_argument("--batch", type=int, default=4, help="per-GPU batch size")
     p.add_argument("--seq", type=int, default=8192)
     p.add_argument("--no-ft", action="store_true",
                    help="disable the fault-tolerance control plane (raw perf)")

@@ -92,10 +92,12 @@ __device__ inline TileRegs load_tiles(const bf16* __restrict__ srcA,
   return r;
 }
 
+template <bool WRITE_TR = true>
 __device__ inline void write_one(unsigned char* rm, unsigned char* tr, int q,
                                  int dseg, uint4 lo, uint4 hi) {
   *reinterpret_cast<uint4*>(rm + rm_addr(q, dseg * 2)) = lo;
   *reinterpret_cast<uint4*>(rm + rm_addr(q, dseg * 2 + 16)) = hi;
+  if constexpr (!WRITE_TR) return;
   // b16 transpose scatter. (A paired-b32 variant using 2 shuffles per
   // element was tried and REGRESSED 17.6 -> 22.0 ms: the VALU shuffle cost
   // exceeds the saved LDS write issue. Next lever is ds_read_b64_tr_b16 on
@@ -109,12 +111,14 @@ __device__ inline void write_one(unsigned char* rm, unsigned char* tr, int q,
   }
 }
 
+template <bool B_TR = true>
 __device__ inline void write_tiles(ImageSet* img, const TileRegs& r) {
   const int t = threadIdx.x;
   const int q = t >> 3;
   const int dseg = (t & 7) * 16;
   write_one(img->a_rm, img->a_tr, q, dseg, r.a_lo, r.a_hi);
-  write_one(img->b_rm, img->b_tr, q, dseg, r.b_lo, r.b_hi);
+  // the dq kernel never reads the transposed V image — skip its scatter
+  write_one<B_TR>(img->b_rm, img->b_tr, q, dseg, r.b_lo, r.b_hi);
 }
 
 __device__ inline bf16x8_vec rm_bfrag(const unsigned char* rm, int t16, int half,
@@ -316,7 +320,7 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
     {
       TileRegs r = load_tiles(k_head + (int64_t)j * FA_BLK * FA_D,
                               v_head + (int64_t)j * FA_BLK * FA_D);
-      write_tiles(&sm.img[cur], r);
+      write_tiles<false>(&sm.img[cur], r);
     }
     __syncthreads();
 

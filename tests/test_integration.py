@@ -48,11 +48,15 @@ class EventInjector:
         return self
 
     def check(self, replica: int, step: int, pg: FakeProcessGroupWrapper) -> None:
-        if self.failures.get(replica) == step:
+        # trigger at >= target: a replica can live-heal PAST the target step
+        # (it jumps to the quorum max_step), which would skip an == check
+        tgt = self.failures.get(replica)
+        if tgt is not None and step >= tgt:
             del self.failures[replica]
             self.count += 1
             raise InjectedFailure(f"injected failure at replica {replica} step {step}")
-        if self.allreduce_failures.get(replica) == step:
+        tgt = self.allreduce_failures.get(replica)
+        if tgt is not None and step >= tgt:
             del self.allreduce_failures[replica]
             self.count += 1
             pg.report_future_error(RuntimeError("injected allreduce failure"))

@@ -220,3 +220,26 @@ class TestDiLoCo:
                 torch.testing.assert_close(a[k], b[k], msg=f"mismatch at {k}")
         finally:
             lh.shutdown()
+
+
+class TestSplitFragments:
+    def test_balanced_split(self):
+        from torchft_amd.local_sgd import split_into_fragments
+
+        model = nn.Sequential(
+            nn.Linear(8, 8), nn.Linear(8, 8), nn.Linear(8, 8), nn.Linear(8, 8)
+        )
+        frags = split_into_fragments(model, 2)
+        assert len(frags) == 2
+        total = sum(p.numel() for p in model.parameters())
+        assert sum(p.numel() for f in frags for p in f.parameters()) == total
+        # fragments alias the original parameters (training flows through)
+        orig_ids = {id(p) for p in model.parameters()}
+        frag_ids = {id(p) for f in frags for p in f.parameters()}
+        assert orig_ids == frag_ids
+
+    def test_too_many_fragments_raises(self):
+        from torchft_amd.local_sgd import split_into_fragments
+
+        with pytest.raises(ValueError):
+            split_into_fragments(nn.Sequential(nn.Linear(2, 2)), 3)

@@ -34,9 +34,22 @@ __all__ = [
     "WorldSizeMode",
 ]
 
-try:  # LocalSGD/DiLoCo are available once local_sgd is importable
-    from torchft_amd.local_sgd import DiLoCo, LocalSGD  # noqa: F401
+from torchft_amd.baby_process_group import (  # noqa: E402
+    ProcessGroupBabyGloo,
+    ProcessGroupBabyRCCL,
+)
+from torchft_amd.local_sgd import DiLoCo, LocalSGD, split_into_fragments  # noqa: E402
+from torchft_amd.process_group import (  # noqa: E402
+    ErrorSwallowingProcessGroupWrapper,
+    ManagedProcessGroup,
+)
 
-    __all__ += ["DiLoCo", "LocalSGD"]
-except ImportError:  # pragma: no cover - during incremental builds
-    pass
+__all__ += [
+    "DiLoCo",
+    "ErrorSwallowingProcessGroupWrapper",
+    "LocalSGD",
+    "ManagedProcessGroup",
+    "ProcessGroupBabyGloo",
+    "ProcessGroupBabyRCCL",
+    "split_into_fragments",
+]

@@ -42,6 +42,38 @@ logger: logging.Logger = logging.getLogger(__name__)
 USE_BUCKETIZATION_ENV: str = "TORCHFT_USE_BUCKETIZATION"
 
 
+def split_into_fragments(model: nn.Module, num_fragments: int) -> List[nn.Module]:
+    """Split a model into ``num_fragments`` pipeline-style fragments for
+    Streaming DiLoCo (the reference cuts with torch.distributed.pipelining;
+    here we cut by top-level children, balancing by parameter count).
+
+    The fragments hold references to the original modules — training the
+    model trains the fragments.
+    """
+    children = [m for m in model.children() if sum(p.numel() for p in m.parameters()) > 0]
+    if len(children) < num_fragments:
+        raise ValueError(
+            f"model has {len(children)} parameterized top-level children, "
+            f"cannot split into {num_fragments} fragments"
+        )
+    total = sum(p.numel() for m in children for p in m.parameters())
+    target = total / num_fragments
+    fragments: List[nn.Module] = []
+    bucket: List[nn.Module] = []
+    acc = 0
+    for i, m in enumerate(children):
+        bucket.append(m)
+        acc += sum(p.numel() for p in m.parameters())
+        remaining_needed = num_fragments - len(fragments) - 1
+        if (acc >= target and remaining_needed > 0 and
+                len(children) - i - 1 >= remaining_needed):
+            fragments.append(nn.Sequential(*bucket) if len(bucket) > 1 else bucket[0])
+            bucket, acc = [], 0
+    fragments.append(nn.Sequential(*bucket) if len(bucket) > 1 else bucket[0])
+    assert len(fragments) == num_fragments
+    return fragments
+
+
 def extract_local_tensor(t: torch.Tensor) -> torch.Tensor:
     """Cloned local representation of a (D)Tensor, detached from grads."""
     if DTensor is not None and isinstance(t, DTensor):

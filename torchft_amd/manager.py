@@ -434,6 +434,7 @@ class Manager:
         assert self._quorum_future is not None, "must call start_quorum before wait_quorum"
         self._quorum_future.result()
 
+    @torch.profiler.record_function("torchft_amd::manager::_async_quorum")
     def _async_quorum(
         self,
         allow_heal: bool,
@@ -507,16 +508,19 @@ class Manager:
                 # RCCL comm abort + re-init; must not race in-flight work.
                 if torch.cuda.is_available():
                     torch.cuda.synchronize()
-                self._pg.configure(
-                    store_prefixed_addr,
-                    self._replica_id if self._replica_id is not None else "0",
-                    replica_rank,
-                    replica_world_size,
-                    quorum_id,
-                    self._group_rank,
-                    self._group_world_size,
-                    ranks_in_quorum,
-                )
+                with torch.profiler.record_function(
+                    "torchft_amd::manager::pg::configure"
+                ):
+                    self._pg.configure(
+                        store_prefixed_addr,
+                        self._replica_id if self._replica_id is not None else "0",
+                        replica_rank,
+                        replica_world_size,
+                        quorum_id,
+                        self._group_rank,
+                        self._group_world_size,
+                        ranks_in_quorum,
+                    )
             except Exception as e:  # noqa: BLE001
                 self._logger.exception(f"got exception in pg configure: {e}")
                 self.report_error(e)
@@ -531,7 +535,10 @@ class Manager:
                         self._logger.info(
                             f"peers need recovery from us {quorum.recover_dst_replica_ranks}"
                         )
-                        self._checkpoint_transport.send_checkpoint(
+                        with torch.profiler.record_function(
+                            "torchft_amd::manager::send_checkpoint"
+                        ):
+                            self._checkpoint_transport.send_checkpoint(
                             dst_ranks=quorum.recover_dst_replica_ranks,
                             step=max_step,
                             state_dict=self._manager_state_dict(),

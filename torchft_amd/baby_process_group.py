@@ -21,11 +21,10 @@ import os
 import queue
 import threading
 from datetime import timedelta
-from typing import Callable, Dict, List, Optional, Union
+from typing import Dict, List, Optional, Union
 
 import torch
 import torch.multiprocessing as mp
-from torch.distributed import PrefixStore, TCPStore
 from torch.distributed.distributed_c10d import (
     AllgatherOptions,
     AllreduceCoalescedOptions,
@@ -40,7 +39,7 @@ from torch.distributed.distributed_c10d import (
 from torch.futures import Future
 
 from torchft_amd.multiprocessing_util import _MonitoredPipe
-from torchft_amd.process_group import ProcessGroup, create_store_client
+from torchft_amd.process_group import ProcessGroup
 
 logger = logging.getLogger(__name__)
 

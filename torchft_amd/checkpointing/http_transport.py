@@ -17,7 +17,6 @@ from __future__ import annotations
 import logging
 import pickle
 import socket
-import struct
 import threading
 import urllib.request
 from concurrent.futures import ThreadPoolExecutor
@@ -32,14 +31,11 @@ from torchft_amd.checkpointing._rwlock import RWLock
 from torchft_amd.checkpointing._serialization import (
     _DTensorMeta,
     _PickledLeaf,
-    _TensorMeta,
     _readinto_tensor,
-    _read_exact,
     _tensor_bytes,
     save_plan,
     split_state_dict,
     streaming_load,
-    streaming_save,
 )
 from torchft_amd.checkpointing.transport import CheckpointTransport
 from torchft_amd.utils import get_stream_context

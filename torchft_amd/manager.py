@@ -24,17 +24,16 @@ import traceback
 import uuid
 import weakref
 from concurrent.futures import ThreadPoolExecutor
-from contextlib import nullcontext
 from datetime import timedelta
 from enum import Enum
-from typing import TYPE_CHECKING, Callable, Dict, List, Optional, TypeVar, cast
+from typing import TYPE_CHECKING, Callable, Dict, Optional, TypeVar, cast
 
 import torch
 import torch.distributed as dist
 from torch.distributed import ReduceOp, TCPStore
 from torch.distributed.distributed_c10d import AllreduceOptions, Work
 
-from torchft_amd._ftcore import LighthouseClient, ManagerClient, ManagerServer
+from torchft_amd._ftcore import ManagerClient, ManagerServer
 from torchft_amd.checkpointing import CheckpointTransport, HTTPTransport
 from torchft_amd.checkpointing._rwlock import RWLock
 from torchft_amd.futures import future_timeout

@@ -291,3 +291,61 @@ class TestEndToEnd:
             assert "Lighthouse" in body
         finally:
             lh.shutdown()
+
+
+class TestResilienceCoord:
+    def test_lighthouse_restart_heartbeat_reconnects(self):
+        """The manager's heartbeat/quorum clients reconnect with backoff
+        when the lighthouse restarts (reference: create_lighthouse_client
+        retry path)."""
+        import time
+
+        lh = core.LighthouseServer(bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=10)
+        port = int(lh.address().rsplit(":", 1)[1])
+        mgr = core.ManagerServer(
+            replica_id="restart0",
+            lighthouse_addr=lh.address(),
+            hostname="127.0.0.1",
+            bind="127.0.0.1:0",
+            store_addr="s",
+            world_size=1,
+            heartbeat_interval=TD(milliseconds=50),
+            connect_timeout=TD(seconds=2),
+        )
+        try:
+            c = core.ManagerClient(mgr.address(), connect_timeout=TD(seconds=5))
+            r = c._quorum(0, 0, "", False, TD(seconds=5))
+            assert r.quorum_id >= 1
+
+            # kill the lighthouse, then bring a new one up on the SAME port
+            lh.shutdown()
+            time.sleep(0.3)
+            lh = core.LighthouseServer(
+                bind=f"127.0.0.1:{port}", min_replicas=1, join_timeout_ms=10
+            )
+            # the manager must re-reach the new lighthouse and form a quorum
+            r = c._quorum(0, 1, "", False, TD(seconds=10))
+            assert r.max_step >= 0
+        finally:
+            mgr.shutdown()
+            lh.shutdown()
+
+    def test_quorum_times_out_when_rank_missing(self):
+        lh = core.LighthouseServer(bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=10)
+        mgr = core.ManagerServer(
+            replica_id="missing0",
+            lighthouse_addr=lh.address(),
+            hostname="127.0.0.1",
+            bind="127.0.0.1:0",
+            store_addr="s",
+            world_size=2,  # second rank never joins
+            heartbeat_interval=TD(milliseconds=50),
+            connect_timeout=TD(seconds=2),
+        )
+        try:
+            c = core.ManagerClient(mgr.address(), connect_timeout=TD(seconds=5))
+            with pytest.raises(TimeoutError):
+                c._quorum(0, 0, "", False, TD(milliseconds=500))
+        finally:
+            mgr.shutdown()
+            lh.shutdown()

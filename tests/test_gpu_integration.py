@@ -150,3 +150,25 @@ class TestCheckpointOnGPU:
         finally:
             src.shutdown()
             dst.shutdown()
+
+
+class TestBabyRCCLOnGPU:
+    def test_world1_allreduce_through_subprocess(self):
+        """RCCL communicator in a child process with a HIP tensor shared via
+        dmabuf IPC (HSA_ENABLE_IPC_MODE_LEGACY=0)."""
+        from torchft_amd.baby_process_group import ProcessGroupBabyRCCL
+
+        store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+        pg = ProcessGroupBabyRCCL(timeout=120.0)
+        pg.configure(f"127.0.0.1:{store.port}/babyrccl", "r0", 0, 1)
+        try:
+            from torch.distributed.distributed_c10d import AllreduceOptions, ReduceOp
+
+            t = torch.full((1024,), 3.0, device="cuda:0")
+            opts = AllreduceOptions()
+            opts.reduceOp = ReduceOp.SUM
+            pg.allreduce([t], opts).wait()
+            torch.cuda.synchronize()
+            torch.testing.assert_close(t, torch.full((1024,), 3.0, device="cuda:0"))
+        finally:
+            pg.shutdown()

@@ -269,3 +269,60 @@ class TestManagerQuorum:
             assert m.errored() is not None
         finally:
             m.shutdown(wait=False)
+
+
+class TestManagedWorkChain:
+    def test_lazy_then_chain_materializes_on_wait(self):
+        from torchft_amd.manager import _ManagedWork
+        from torchft_amd.work import _DummyWork
+
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum()
+        m = make_manager(client)
+        try:
+            m.start_quorum()
+            m.wait_quorum()
+            t = torch.ones(4)
+            inner = _DummyWork([t])
+            work = _ManagedWork(m, inner, t)
+            fut = work.get_future()
+            calls = []
+
+            def cb(f):
+                calls.append(1)
+                return f.value() * 2
+
+            fut = fut.then(cb)
+            # then() is lazy: nothing ran yet
+            assert calls == []
+            assert work.wait()
+            assert calls == [1]
+            torch.testing.assert_close(fut.wait(), torch.full((4,), 2.0))
+        finally:
+            m.shutdown(wait=False)
+
+    def test_allreduce_sum_skips_normalization(self):
+        from torch.distributed import ReduceOp
+
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum()
+        m = make_manager(client)
+        try:
+            m.start_quorum()
+            t = torch.ones(4)
+            work = m.allreduce(t, reduce_op=ReduceOp.SUM)
+            assert work.wait()
+            torch.testing.assert_close(t, torch.ones(4))  # no /participants
+        finally:
+            m.shutdown(wait=False)
+
+    def test_avg_rejects_integer_tensors(self):
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum()
+        m = make_manager(client)
+        try:
+            m.start_quorum()
+            with pytest.raises(ValueError, match="floating point"):
+                m.allreduce(torch.ones(4, dtype=torch.int64))
+        finally:
+            m.shutdown(wait=False)

@@ -281,6 +281,7 @@ class _StreamingDiLoCoFragment:
 
         self._manager.register_state_dict_fn(fragment_key, load_fn, save_fn)
 
+    @torch.profiler.record_function("torchft_amd::local_sgd::save_parameters")
     def save_parameters(self) -> None:
         with torch.no_grad():
             for name, p in self._model_fragment.named_parameters():
@@ -292,6 +293,7 @@ class _StreamingDiLoCoFragment:
             for name, p in self._model_fragment.named_parameters():
                 self._local_parameters[name] = extract_local_tensor(p.data)
 
+    @torch.profiler.record_function("torchft_amd::local_sgd::restore_parameters")
     def restore_parameters(self) -> None:
         with torch.no_grad():
             for name, p in self._model_fragment.named_parameters():
@@ -365,6 +367,7 @@ class _StreamingDiLoCoFragment:
             self._stop_event = None
         self._allreduce_work = []
 
+    @torch.profiler.record_function("torchft_amd::local_sgd::prepare_sync")
     def prepare_sync(self) -> None:
         """Compute pseudo-gradients and launch (but don't wait for) the
         allreduce on the side stream."""
@@ -377,6 +380,7 @@ class _StreamingDiLoCoFragment:
         ):
             self._average_grads()
 
+    @torch.profiler.record_function("torchft_amd::local_sgd::perform_sync")
     def perform_sync(self) -> bool:
         """Wait for the allreduce, then commit (outer-step) or roll back."""
         assert len(self._allreduce_work) > 0

@@ -118,7 +118,10 @@ class Lighthouse {
   std::thread tick_thread_;
   std::mutex conns_mu_;
   std::set<int> conns_;
-  std::vector<std::thread> conn_threads_;
+  // connection threads are detached; this counts the live ones so
+  // shutdown() can wait them out (a per-thread vector would grow
+  // unreaped over a long-running job)
+  std::atomic<int> active_conns_{0};
 };
 
 // ------------------------------------------------------------- manager
@@ -137,6 +140,7 @@ class ManagerSrv {
   void accept_loop();
   void heartbeat_loop();
   void handle_conn(int fd);
+  void quorum_worker_loop();
   void run_quorum(QuorumMember member, Millis timeout);
 
   std::string replica_id_;
@@ -169,8 +173,13 @@ class ManagerSrv {
   std::thread heartbeat_thread_;
   std::mutex conns_mu_;
   std::set<int> conns_;
-  std::vector<std::thread> conn_threads_;
-  std::vector<std::thread> quorum_threads_;
+  std::atomic<int> active_conns_{0};
+
+  // single long-lived quorum runner (latest-wins trigger slot): one
+  // std::thread per quorum round would accumulate unreaped over the job
+  std::thread quorum_worker_;
+  std::condition_variable qcv_;
+  std::optional<std::pair<QuorumMember, Millis>> pending_quorum_;
 };
 
 }  // namespace ftcoord

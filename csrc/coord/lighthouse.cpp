@@ -142,13 +142,11 @@ void Lighthouse::shutdown() {
   }
   if (accept_thread_.joinable()) accept_thread_.join();
   if (tick_thread_.joinable()) tick_thread_.join();
-  std::vector<std::thread> threads;
-  {
-    std::lock_guard<std::mutex> g(conns_mu_);
-    threads.swap(conn_threads_);
+  // detached connection threads observe stop_/closed fds and exit; wait
+  // them out (bounded) so no thread touches this object after destruction
+  for (int i = 0; i < 600 && active_conns_.load() > 0; i++) {
+    std::this_thread::sleep_for(Millis(10));
   }
-  for (auto& t : threads)
-    if (t.joinable()) t.join();
 }
 
 void Lighthouse::tick_loop() {
@@ -203,14 +201,20 @@ void Lighthouse::accept_loop() {
     int fd = ::accept(listen_fd_, nullptr, nullptr);
     if (fd < 0) continue;
     set_nodelay(fd);
-    std::lock_guard<std::mutex> g(conns_mu_);
-    conns_.insert(fd);
-    conn_threads_.emplace_back([this, fd] {
+    {
+      std::lock_guard<std::mutex> g(conns_mu_);
+      conns_.insert(fd);
+    }
+    active_conns_.fetch_add(1);
+    std::thread([this, fd] {
       handle_conn(fd);
-      std::lock_guard<std::mutex> g2(conns_mu_);
-      conns_.erase(fd);
+      {
+        std::lock_guard<std::mutex> g2(conns_mu_);
+        conns_.erase(fd);
+      }
       ::close(fd);
-    });
+      active_conns_.fetch_sub(1);
+    }).detach();
   }
 }
 

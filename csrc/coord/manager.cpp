@@ -107,6 +107,7 @@ ManagerSrv::ManagerSrv(std::string replica_id, std::string lighthouse_addr, std:
   }
   accept_thread_ = std::thread([this] { accept_loop(); });
   heartbeat_thread_ = std::thread([this] { heartbeat_loop(); });
+  quorum_worker_ = std::thread([this] { quorum_worker_loop(); });
 }
 
 ManagerSrv::~ManagerSrv() { shutdown(); }
@@ -123,22 +124,17 @@ void ManagerSrv::shutdown() {
     ::close(listen_fd_);
   }
   cv_.notify_all();
+  qcv_.notify_all();
   {
     std::lock_guard<std::mutex> g(conns_mu_);
     for (int fd : conns_) ::shutdown(fd, SHUT_RDWR);
   }
   if (accept_thread_.joinable()) accept_thread_.join();
   if (heartbeat_thread_.joinable()) heartbeat_thread_.join();
-  std::vector<std::thread> threads;
-  {
-    std::lock_guard<std::mutex> g(conns_mu_);
-    threads.swap(conn_threads_);
-    threads.insert(threads.end(), std::make_move_iterator(quorum_threads_.begin()),
-                   std::make_move_iterator(quorum_threads_.end()));
-    quorum_threads_.clear();
+  if (quorum_worker_.joinable()) quorum_worker_.join();
+  for (int i = 0; i < 600 && active_conns_.load() > 0; i++) {
+    std::this_thread::sleep_for(Millis(10));
   }
-  for (auto& t : threads)
-    if (t.joinable()) t.join();
 }
 
 void ManagerSrv::heartbeat_loop() {
@@ -204,14 +200,38 @@ void ManagerSrv::accept_loop() {
     int fd = ::accept(listen_fd_, nullptr, nullptr);
     if (fd < 0) continue;
     set_nodelay(fd);
-    std::lock_guard<std::mutex> g(conns_mu_);
-    conns_.insert(fd);
-    conn_threads_.emplace_back([this, fd] {
+    {
+      std::lock_guard<std::mutex> g(conns_mu_);
+      conns_.insert(fd);
+    }
+    active_conns_.fetch_add(1);
+    std::thread([this, fd] {
       handle_conn(fd);
-      std::lock_guard<std::mutex> g2(conns_mu_);
-      conns_.erase(fd);
+      {
+        std::lock_guard<std::mutex> g2(conns_mu_);
+        conns_.erase(fd);
+      }
       ::close(fd);
-    });
+      active_conns_.fetch_sub(1);
+    }).detach();
+  }
+}
+
+void ManagerSrv::quorum_worker_loop() {
+  // one long-lived runner; handle_conn drops (member, timeout) into the
+  // latest-wins slot when the last rank of a round joins
+  while (!stop_.load()) {
+    std::pair<QuorumMember, Millis> job;
+    {
+      std::unique_lock<std::mutex> lk(mu_);
+      qcv_.wait_for(lk, Millis(250),
+                    [&] { return stop_.load() || pending_quorum_.has_value(); });
+      if (stop_.load()) return;
+      if (!pending_quorum_.has_value()) continue;
+      job = std::move(*pending_quorum_);
+      pending_quorum_.reset();
+    }
+    run_quorum(job.first, job.second);
   }
 }
 
@@ -263,9 +283,8 @@ void ManagerSrv::handle_conn(int fd) {
 
             if ((int64_t)participants_.size() == world_size_) {
               participants_.clear();
-              std::lock_guard<std::mutex> g(conns_mu_);
-              quorum_threads_.emplace_back(
-                  [this, member, timeout_ms] { run_quorum(member, Millis(timeout_ms)); });
+              pending_quorum_ = {member, Millis(timeout_ms)};
+              qcv_.notify_one();
             }
 
             while (!stop_.load() && quorum_seq_ == seen) {

@@ -160,7 +160,18 @@ void Lighthouse::tick_loop() {
 }
 
 void Lighthouse::tick_locked(std::unique_lock<std::mutex>& lk) {
-  auto [met, reason] = quorum_compute(Clock::now(), state_, opt_);
+  // prune long-dead heartbeat entries so a job with many replica
+  // restarts (each a fresh uuid) doesn't grow state without bound
+  const auto now = Clock::now();
+  const auto prune_age = Millis(opt_.heartbeat_timeout_ms * 10);
+  for (auto it = state_.heartbeats.begin(); it != state_.heartbeats.end();) {
+    if (now - it->second > prune_age && !state_.participants.count(it->first)) {
+      it = state_.heartbeats.erase(it);
+    } else {
+      ++it;
+    }
+  }
+  auto [met, reason] = quorum_compute(now, state_, opt_);
   if (reason != last_reason_) {
     last_reason_ = reason;  // change-logged status, mirrors ChangeLogger
   }

@@ -98,3 +98,11 @@ class TestContextParallel:
         torch.testing.assert_close(dq, ref_grads[0], rtol=2e-3, atol=2e-3)
         torch.testing.assert_close(dk, ref_grads[1], rtol=2e-3, atol=2e-3)
         torch.testing.assert_close(dv, ref_grads[2], rtol=2e-3, atol=2e-3)
+
+
+class TestShardGuards:
+    def test_indivisible_sequence_rejected(self):
+        import pytest as _pytest
+
+        with _pytest.raises(AssertionError):
+            shard_sequence(torch.zeros(1, 30, 2, 4), 0, 4)

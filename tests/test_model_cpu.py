@@ -65,3 +65,23 @@ class TestLlamaCPU:
         l1 = m1.forward_loss(toks[:, :-1], toks[:, 1:])
         l2 = m2.forward_loss(toks[:, :-1], toks[:, 1:])
         torch.testing.assert_close(l1, l2)
+
+
+class TestFusedAdamWCPUFallback:
+    def test_matches_torch_adamw(self):
+        from torchft_amd.ops import FusedAdamW
+
+        torch.manual_seed(9)
+        p = torch.randn(200, requires_grad=True)
+        ref = p.detach().clone().requires_grad_(True)
+        opt = FusedAdamW([p], lr=1e-2, betas=(0.9, 0.95), weight_decay=0.01)
+        ref_opt = torch.optim.AdamW(
+            [ref], lr=1e-2, betas=(0.9, 0.95), weight_decay=0.01, eps=1e-8
+        )
+        for _ in range(5):
+            g = torch.randn_like(p)
+            p.grad = g.clone()
+            ref.grad = g.clone()
+            opt.step()
+            ref_opt.step()
+        torch.testing.assert_close(p, ref, rtol=1e-5, atol=1e-6)

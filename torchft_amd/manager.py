@@ -538,11 +538,11 @@ class Manager:
                             "torchft_amd::manager::send_checkpoint"
                         ):
                             self._checkpoint_transport.send_checkpoint(
-                            dst_ranks=quorum.recover_dst_replica_ranks,
-                            step=max_step,
-                            state_dict=self._manager_state_dict(),
-                            timeout=self._timeout,
-                        )
+                                dst_ranks=quorum.recover_dst_replica_ranks,
+                                step=max_step,
+                                state_dict=self._manager_state_dict(),
+                                timeout=self._timeout,
+                            )
 
                     if heal:
                         self._healing = True

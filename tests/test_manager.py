@@ -326,3 +326,17 @@ class TestManagedWorkChain:
                 m.allreduce(torch.ones(4, dtype=torch.int64))
         finally:
             m.shutdown(wait=False)
+
+
+class TestQuorumArgsPropagation:
+    def test_shrink_only_forwarded(self):
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum()
+        m = make_manager(client)
+        try:
+            m.start_quorum(shrink_only=True)
+            m.wait_quorum()
+            assert client._quorum.call_args.kwargs["shrink_only"] is True
+            assert client._quorum.call_args.kwargs["init_sync"] is True
+        finally:
+            m.shutdown(wait=False)

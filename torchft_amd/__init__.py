@@ -7,6 +7,18 @@ fp8-quantized collectives and fused optimizer steps, and HIP-stream-based
 live healing.
 """
 
+# The C++ coordination core is built in-tree. On a fresh checkout it does
+# not exist yet — build it on first import (g++ only, a few seconds) so
+# `import torchft_amd` and `python -m torchft_amd._build` both work from
+# pristine state. The HIP kernel extension stays lazy (torchft_amd.ops).
+try:
+    from torchft_amd import _ftcore  # noqa: F401
+except ImportError:  # pragma: no cover - fresh checkout
+    from torchft_amd._build import build_ftcore as _build_ftcore
+
+    _build_ftcore()
+    from torchft_amd import _ftcore  # noqa: F401
+
 from torchft_amd.data import DistributedSampler
 from torchft_amd.ddp import DistributedDataParallel
 from torchft_amd.manager import Manager, WorldSizeMode

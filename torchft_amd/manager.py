@@ -1,32 +1,46 @@
 """Per-rank fault-tolerant training control plane for MI355X.
 
-The ``Manager`` drives the per-step protocol: (async) quorum formation
-against the C++ lighthouse/manager services, RCCL communicator
-reconfiguration on membership change, live healing of joined/behind replicas
-over a checkpoint transport on a dedicated HIP recovery stream, error
-tracking, and the all-ranks ``should_commit`` barrier that gates every
-optimizer step.
+The ``Manager`` runs an explicit per-step protocol:
 
-Reference parity (semantics): torchft/manager.py — ctor (:162-364),
-``allreduce`` (:410-493), ``report_error/errored`` (:495-514),
-``wrap_future`` (:516-558), ``start_quorum``/``_async_quorum`` (:560-813),
-``should_commit`` (:855-943), state dict registry (:380-399),
-``_ManagedWork``/``_ManagedFuture`` lazy callback chain (:1080-1363).
+    start_quorum()  -> negotiate      (quorum RPC against the C++ services,
+                                       on a background worker thread)
+                    -> participate    (pure function: quorum -> who trains)
+                    -> rebuild comm   (RCCL abort + re-init when membership
+                                       changed; per-quorum store prefix)
+                    -> recover        (serve or fetch a live checkpoint on a
+                                       dedicated HIP recovery stream)
+    allreduce()     -> fault-tolerant gradient averaging, continuations run
+                       on the issuing HIP stream (``_ManagedWork``)
+    should_commit() -> drain recovery + compute streams, apply staged heal,
+                       all-ranks barrier; only then may the optimizer step.
+
+Architecture notes (deliberately different from the reference,
+torchft/manager.py, whose semantics we match — ctor :162-364, allreduce
+:410-493, quorum :560-813, should_commit :855-943, managed work :1080-1363):
+
+* each ``start_quorum`` produces a ``_StepAttempt`` record; participation is
+  computed by the pure function ``_participation_from`` (unit-testable
+  without any I/O);
+* the quorum cycle is a pipeline of small named methods instead of one
+  monolithic thread function;
+* ``_ManagedWork`` keeps a flat list of continuations and materializes them
+  with a single real ``Future.then`` at wait time, bound to the issuing
+  HIP stream — not a linked list of lazy futures.
 """
 
 from __future__ import annotations
 
-import concurrent.futures
 import logging
 import os
+import queue
 import socket
+import threading
 import traceback
 import uuid
-import weakref
-from concurrent.futures import ThreadPoolExecutor
+from dataclasses import dataclass
 from datetime import timedelta
 from enum import Enum
-from typing import TYPE_CHECKING, Callable, Dict, Optional, TypeVar, cast
+from typing import TYPE_CHECKING, Callable, Dict, List, Optional, Tuple, TypeVar, cast
 
 import torch
 import torch.distributed as dist
@@ -44,7 +58,6 @@ if TYPE_CHECKING:
     from torchft_amd.process_group import ProcessGroup
 
 T = TypeVar("T")
-S = TypeVar("S")
 
 logger: logging.Logger = logging.getLogger(__name__)
 
@@ -59,26 +72,26 @@ QUORUM_RETRIES_ENV: str = "TORCHFT_QUORUM_RETRIES"
 
 
 def get_timeout(timeout_sec_env: Optional[str], default: timedelta) -> timedelta:
-    if timeout_sec_env is not None:
-        return timedelta(seconds=int(timeout_sec_env))
-    return default
+    """Env-var seconds override for a timedelta default."""
+    return timedelta(seconds=int(timeout_sec_env)) if timeout_sec_env else default
 
 
 def extract_trailing_digits(s: str) -> int:
-    """Extract trailing digits of a string, e.g. ``"replica_5"`` → 5."""
-    i = len(s)
-    while i > 0 and s[i - 1].isdigit():
-        i -= 1
-    return int(s[i:]) if i < len(s) else 0
+    """``"replica_57"`` -> 57; no trailing digits -> 0."""
+    digits = ""
+    for ch in reversed(s):
+        if not ch.isdigit():
+            break
+        digits = ch + digits
+    return int(digits) if digits else 0
 
 
 class WorldSizeMode(Enum):
-    """How the world size is handled when replicas join/leave.
+    """Numeric handling of a changing replica count.
 
-    DYNAMIC: the world size may change between steps; batch size (and thus
-        learning dynamics) varies with membership.
-    FIXED_WITH_SPARES: at most ``min_replica_size`` replicas participate
-        numerically; extras run but their gradients are discarded.
+    DYNAMIC: effective batch size follows membership.
+    FIXED_WITH_SPARES: at most ``min_replica_size`` replicas contribute
+        gradients; the rest train but are numerically inert spares.
     """
 
     DYNAMIC = 0
@@ -92,14 +105,136 @@ class ExceptionWithTraceback(Exception):
         super().__init__(f"{e}\n{self.stack_trace}")
 
 
+# ---------------------------------------------------------------------------
+# background worker
+# ---------------------------------------------------------------------------
+
+
+class _Pending:
+    """Completion handle for a task submitted to ``_SerialExecutor``."""
+
+    def __init__(self) -> None:
+        self._done = threading.Event()
+        self._exc: Optional[BaseException] = None
+
+    def _finish(self, exc: Optional[BaseException]) -> None:
+        self._exc = exc
+        self._done.set()
+
+    def result(self, timeout: Optional[float] = None) -> None:
+        if not self._done.wait(timeout):
+            raise TimeoutError("quorum task did not complete in time")
+        if self._exc is not None:
+            raise self._exc
+
+
+class _SerialExecutor:
+    """One daemon thread executing submitted thunks strictly in order.
+
+    The quorum negotiation must never run concurrently with itself and must
+    not block the training thread; a dedicated serial worker makes both
+    properties structural instead of incidental.
+    """
+
+    def __init__(self, name: str) -> None:
+        self._q: "queue.Queue[Optional[Tuple[Callable[[], None], _Pending]]]" = (
+            queue.Queue()
+        )
+        self._thread = threading.Thread(target=self._run, name=name, daemon=True)
+        self._started = False
+
+    def submit(self, fn: Callable[[], None]) -> _Pending:
+        if not self._started:
+            self._thread.start()
+            self._started = True
+        pending = _Pending()
+        self._q.put((fn, pending))
+        return pending
+
+    def _run(self) -> None:
+        while True:
+            item = self._q.get()
+            if item is None:
+                return
+            fn, pending = item
+            try:
+                fn()
+                pending._finish(None)
+            except BaseException as e:  # noqa: BLE001
+                pending._finish(e)
+
+    def shutdown(self, wait: bool = True) -> None:
+        if not self._started:
+            return
+        self._q.put(None)
+        if wait:
+            self._thread.join(timeout=60)
+
+
+# ---------------------------------------------------------------------------
+# per-step state
+# ---------------------------------------------------------------------------
+
+
+@dataclass
+class _StepAttempt:
+    """Everything the current step learned from its quorum."""
+
+    pending: Optional[_Pending] = None
+    # filled in by the quorum pipeline:
+    participant_rank: Optional[int] = None
+    participant_count: int = 0
+    healing: bool = False
+    staged_state: Optional[Dict[str, object]] = None
+    recovery_event: Optional[torch.cuda.Event] = None
+    error: Optional[ExceptionWithTraceback] = None
+
+    def ready(self) -> None:
+        assert self.pending is not None, "start_quorum must be called first"
+        self.pending.result()
+
+
+def _participation_from(
+    *,
+    replica_rank: int,
+    replica_world_size: int,
+    max_replica_rank: Optional[int],
+    max_world_size: int,
+    defer_healing: bool,
+    world_size_mode: WorldSizeMode,
+    min_replica_size: int,
+) -> Tuple[Optional[int], int]:
+    """Pure function: quorum result -> (participating rank | None, count).
+
+    With deferred healing (async quorum) only the replicas already at
+    max_step train this step; otherwise everyone in the quorum does.
+    FIXED_WITH_SPARES then caps the numeric world at ``min_replica_size``
+    and benches the overflow ranks.
+    """
+    if defer_healing:
+        rank, count = max_replica_rank, max_world_size
+    else:
+        rank, count = replica_rank, replica_world_size
+
+    if world_size_mode == WorldSizeMode.FIXED_WITH_SPARES:
+        count = min(count, min_replica_size)
+        if rank is not None and rank >= min_replica_size:
+            rank = None
+    return rank, count
+
+
+# ---------------------------------------------------------------------------
+# the Manager
+# ---------------------------------------------------------------------------
+
+
 class Manager:
-    """Fault-tolerant training loop manager (one per rank).
+    """Fault-tolerant training loop manager (one instance per rank).
 
-    Requires the replica group's TCPStore (MASTER_ADDR/PORT or
-    store_addr/store_port) to already exist — torchrun provides it.
-
-    NOTE: when saving periodic checkpoints you must save/restore the
-    Manager's ``state_dict`` as well to avoid synchronization issues.
+    Expects the replica group's TCPStore to exist already (torchrun creates
+    it; MASTER_ADDR/PORT or store_addr/store_port point at it). When saving
+    periodic checkpoints, persist ``Manager.state_dict()`` alongside the
+    model so a restore resumes the step/batch counters consistently.
     """
 
     def __init__(
@@ -128,23 +263,19 @@ class Manager:
         quorum_retries: int = 0,
         should_quantize: bool = False,
     ) -> None:
+        # telemetry channels (structured records; see telemetry.py)
         self.quorum_logger: logging.Logger = logging.getLogger("torchft_quorums")
         self.commits_logger: logging.Logger = logging.getLogger("torchft_commits")
         self.errors_logger: logging.Logger = logging.getLogger("torchft_errors")
 
-        self._load_state_dict_fns: Dict[str, Callable[[object], None]] = {}
-        self._user_state_dicts: Dict[str, Callable[[], object]] = {}
-
-        self._replica_id = replica_id
-
-        # Guards state-dict reads (checkpoint serving) vs training mutation.
-        self._state_dict_lock = RWLock(timeout=timeout.total_seconds())
-
-        if load_state_dict and state_dict:
-            self.register_state_dict_fn("default", load_state_dict, state_dict)
-
-        self._pending_state_dict: Optional[Dict[str, object]] = None
+        self._pg = pg
+        self._min_replica_size = min_replica_size
         self._use_async_quorum = use_async_quorum
+        self._world_size_mode = world_size_mode
+        self._init_sync = init_sync
+        self._max_retries = max_retries
+        self._default_should_quantize = should_quantize
+        self._replica_id = replica_id
 
         self._timeout = get_timeout(os.environ.get(TIMEOUT_SEC_ENV), timeout)
         self._quorum_timeout = get_timeout(
@@ -153,23 +284,32 @@ class Manager:
         self._connect_timeout = get_timeout(
             os.environ.get(CONNECT_TIMEOUT_SEC_ENV), connect_timeout
         )
-
-        self._default_should_quantize = should_quantize
-        self._replica_world_size_mode = world_size_mode
-        self._init_sync = init_sync
-        self._max_retries = max_retries
-        self._commit_failures = 0
         self._quorum_retries: int = int(
             os.environ.get(QUORUM_RETRIES_ENV, str(quorum_retries))
         )
 
-        store_addr = store_addr or os.environ["MASTER_ADDR"]
-        store_port = store_port or int(os.environ["MASTER_PORT"])
         self._group_rank: int = rank if rank is not None else int(os.environ["RANK"])
-        group_rank = self._group_rank
         self._group_world_size: int = world_size or int(os.environ["WORLD_SIZE"])
-        self._min_replica_size = min_replica_size
 
+        # ---- state-dict registry + serving lock --------------------------
+        self._load_state_dict_fns: Dict[str, Callable[[object], None]] = {}
+        self._user_state_dicts: Dict[str, Callable[[], object]] = {}
+        self._state_dict_lock = RWLock(timeout=timeout.total_seconds())
+        self._is_state_dict_read_allowed = True
+        if load_state_dict and state_dict:
+            self.register_state_dict_fn("default", load_state_dict, state_dict)
+
+        # ---- durable protocol counters ------------------------------------
+        self._step = 0
+        self._batches_committed = 0
+        self._commit_failures = 0
+        self._quorum_id = -1
+
+        # ---- per-step state ----------------------------------------------
+        self._attempt = _StepAttempt()
+        self._attempt.pending = None
+
+        # ---- transports ---------------------------------------------------
         if checkpoint_transport is None:
             checkpoint_transport = HTTPTransport[Dict[str, T]](
                 timeout=timeout, num_chunks=0
@@ -178,26 +318,23 @@ class Manager:
             checkpoint_transport
         )
 
-        self._executor = ThreadPoolExecutor(
-            max_workers=1, thread_name_prefix="async_quorum"
+        # Healing copies run on their own HIP stream so a multi-GB fetch
+        # overlaps compute; should_commit waits on the recorded event.
+        self._recovery_stream: Optional[torch.cuda.Stream] = (
+            torch.cuda.Stream() if torch.cuda.is_available() else None
         )
-        self._quorum_future: Optional[concurrent.futures.Future] = None
 
+        self._worker = _SerialExecutor("torchft-quorum")
+
+        # ---- coordination services ---------------------------------------
+        store_addr = store_addr or os.environ["MASTER_ADDR"]
+        store_port = store_port or int(os.environ["MASTER_PORT"])
         self._store = TCPStore(
             host_name=store_addr,
             port=store_port,
             is_master=False,
             wait_for_workers=False,
         )
-        self._pg = pg
-        self._manager: Optional[ManagerServer] = None
-
-        # Healing runs on its own HIP stream so checkpoint send/recv overlap
-        # with compute; should_commit syncs on the recorded recovery event.
-        self._recovery_stream: Optional[torch.cuda.Stream] = (
-            torch.cuda.Stream() if torch.cuda.is_available() else None
-        )
-        self._recovery_event: Optional[torch.cuda.Event] = None
 
         if hostname is None:
             hostname = socket.gethostname()
@@ -206,21 +343,22 @@ class Manager:
             except socket.gaierror:
                 hostname = "127.0.0.1"
 
+        self._manager: Optional[ManagerServer] = None
         if self._group_rank == 0:
+            # group rank 0 hosts the per-replica aggregation service and
+            # publishes its address through the group store
             if port is None:
                 port = int(os.environ.get(MANAGER_PORT_ENV, 0))
-            bind = f"0.0.0.0:{port}"
-            lighthouse_addr = lighthouse_addr or os.environ[LIGHTHOUSE_ADDR_ENV]
-
-            # Unique suffix so a fast restart of the same replica name is a
-            # distinct member at the lighthouse.
-            new_uuid = str(uuid.uuid4())
-            replica_id = new_uuid if not replica_id else f"{replica_id}:{new_uuid}"
+            effective_id = str(uuid.uuid4())
+            if replica_id:
+                # unique suffix: a fast restart of the same replica name must
+                # register as a distinct member at the lighthouse
+                effective_id = f"{replica_id}:{effective_id}"
             self._manager = ManagerServer(
-                replica_id=replica_id,
-                lighthouse_addr=lighthouse_addr,
+                replica_id=effective_id,
+                lighthouse_addr=lighthouse_addr or os.environ[LIGHTHOUSE_ADDR_ENV],
                 hostname=hostname,
-                bind=bind,
+                bind=f"0.0.0.0:{port}",
                 store_addr=f"{store_addr}:{store_port}",
                 world_size=self._group_world_size,
                 heartbeat_interval=heartbeat_interval,
@@ -228,48 +366,24 @@ class Manager:
                 quorum_retries=self._quorum_retries,
             )
             self._store.set(MANAGER_ADDR_KEY, self._manager.address())
-            self._store.set(REPLICA_ID_KEY, replica_id)
+            self._store.set(REPLICA_ID_KEY, effective_id)
 
-        addr = self._store.get(MANAGER_ADDR_KEY).decode("utf-8")
-        self._client = ManagerClient(addr, connect_timeout=connect_timeout)
+        manager_addr = self._store.get(MANAGER_ADDR_KEY).decode("utf-8")
+        self._client = ManagerClient(manager_addr, connect_timeout=connect_timeout)
 
-        replica_id = self._store.get(REPLICA_ID_KEY).decode("utf-8")
+        published_id = self._store.get(REPLICA_ID_KEY).decode("utf-8")
         self._logger = _ManagerLogger(
-            manager=self, replica_id=replica_id or "", group_rank=group_rank
+            manager=self, replica_id=published_id or "", group_rank=self._group_rank
         )
-
-        self._step = 0
-        self._quorum_id = -1
-        self._errored: Optional[ExceptionWithTraceback] = None
-        self._healing = False
-        self._batches_committed = 0
-
-        self._participating_replica_rank: Optional[int] = None
-        self._participating_replica_world_size: int = 0
-        self._is_state_dict_read_allowed = True
 
         self._global_rank: int = (
             self._group_rank
             if self._replica_id is None
-            else (
-                extract_trailing_digits(self._replica_id) * self._group_world_size
-                + self._group_rank
-            )
+            else extract_trailing_digits(self._replica_id) * self._group_world_size
+            + self._group_rank
         )
 
-    # -- state-dict registry ------------------------------------------------
-
-    def allow_state_dict_read(self) -> None:
-        if self._is_state_dict_read_allowed:
-            return
-        self._is_state_dict_read_allowed = True
-        self._state_dict_lock.w_release()
-
-    def disallow_state_dict_read(self) -> None:
-        if not self._is_state_dict_read_allowed:
-            return
-        self._is_state_dict_read_allowed = False
-        self._state_dict_lock.w_acquire()
+    # -- state-dict registry --------------------------------------------------
 
     def register_state_dict_fn(
         self,
@@ -290,13 +404,23 @@ class Manager:
         )
         self.register_state_dict_fn("set_state_dict_fns", load_state_dict, state_dict)
 
+    def allow_state_dict_read(self) -> None:
+        if not self._is_state_dict_read_allowed:
+            self._is_state_dict_read_allowed = True
+            self._state_dict_lock.w_release()
+
+    def disallow_state_dict_read(self) -> None:
+        if self._is_state_dict_read_allowed:
+            self._is_state_dict_read_allowed = False
+            self._state_dict_lock.w_acquire()
+
     def shutdown(self, wait: bool = True) -> None:
         self._checkpoint_transport.shutdown(wait=wait)
         if self._manager is not None:
             self._manager.shutdown()
-        self._executor.shutdown(wait=wait)
+        self._worker.shutdown(wait=wait)
 
-    # -- the hot path --------------------------------------------------------
+    # -- hot path: fault-tolerant allreduce -----------------------------------
 
     @torch.profiler.record_function("torchft_amd::manager::allreduce")
     def allreduce(
@@ -305,14 +429,16 @@ class Manager:
         should_quantize: Optional[bool] = None,
         reduce_op: ReduceOp = ReduceOp.AVG,
     ) -> Work:
-        """Fault-tolerant allreduce; AVG scales by 1/num_participants.
+        """Fault-tolerant allreduce of ``tensor``.
 
-        On error the returned work completes successfully with the tensor
-        untouched (and possibly corrupted — zero before reuse); the error is
-        tracked and surfaces at ``should_commit``.
+        AVG divides by the live participant count as a continuation on the
+        issuing stream. On any error the returned work still "succeeds"
+        (tensor contents undefined — zero before reuse); the error is
+        remembered and fails the step at ``should_commit``.
 
-        ``should_quantize=True`` routes through the fp8 quantized allreduce
-        (CDNA4 HIP kernels + alltoall/allgather over all 7 xGMI links).
+        ``should_quantize=True`` takes the fp8 path: CDNA4 quantize kernels
+        plus an alltoall/allgather decomposition that engages all 7 xGMI
+        links.
         """
         if should_quantize is None:
             should_quantize = self._default_should_quantize
@@ -321,53 +447,49 @@ class Manager:
             return _DummyWork(tensor)
 
         self.wait_quorum()
-        num_participants: int = self.num_participants()
+        num_participants = self.num_participants()
 
         if not self.is_participating():
             tensor.zero_()
 
-        pg_reduce_op = reduce_op
-        if reduce_op == ReduceOp.AVG:
-            if not torch.is_floating_point(tensor):
-                raise ValueError(
-                    "average reduce op is only supported for floating point tensors"
-                )
-            pg_reduce_op = ReduceOp.SUM
+        if reduce_op == ReduceOp.AVG and not torch.is_floating_point(tensor):
+            raise ValueError(
+                "average reduce op is only supported for floating point tensors"
+            )
+        wire_op = ReduceOp.SUM if reduce_op == ReduceOp.AVG else reduce_op
 
         try:
             if should_quantize and torch.cuda.is_available():
                 from torchft_amd.collectives import allreduce_quantized
 
                 work = allreduce_quantized(
-                    [tensor], pg_reduce_op, self._pg, torch.cuda.current_stream()
+                    [tensor], wire_op, self._pg, torch.cuda.current_stream()
                 )
             else:
                 opts = AllreduceOptions()
-                opts.reduceOp = pg_reduce_op
+                opts.reduceOp = wire_op
                 work = self._pg.allreduce([tensor], opts)
-
-            # grad normalization as a continuation on the future chain
-            def callback(fut: torch.futures.Future[torch.Tensor]) -> torch.Tensor:
-                nonlocal tensor
-                if reduce_op == ReduceOp.AVG:
-                    tensor /= num_participants
-                return tensor
-
-            managed_work = _ManagedWork(self, work, tensor)
-            fut = cast(torch.futures.Future[torch.Tensor], managed_work.get_future())
-            fut = fut.then(callback)
-            return managed_work
         except Exception as e:  # noqa: BLE001
-            self._logger.exception(f"got exception in all reduce -- skipping remaining: {e}")
+            self._logger.exception(f"allreduce failed to launch: {e}")
             self.report_error(e)
             return _DummyWork(tensor)
 
+        managed = _ManagedWork(self, work, tensor)
+        if reduce_op == ReduceOp.AVG:
+
+            def normalize(value: torch.Tensor) -> torch.Tensor:
+                value /= num_participants
+                return value
+
+            managed._append(normalize)
+        return managed
+
     def report_error(self, e: Exception) -> None:
-        """Mark this step as failed (gradients must be discarded)."""
-        self._errored = ExceptionWithTraceback(e)
+        """Mark the in-flight step failed; its gradients must be discarded."""
+        self._attempt.error = ExceptionWithTraceback(e)
 
     def errored(self) -> Optional[ExceptionWithTraceback]:
-        return self._errored
+        return self._attempt.error
 
     def wrap_future(
         self,
@@ -375,27 +497,23 @@ class Manager:
         default: T,
         timeout: Optional[timedelta] = None,
     ) -> torch.futures.Future[T]:
-        """Swallow errors on ``fut``, report them, and complete with
-        ``default`` instead; also applies a timeout."""
-        fut = future_timeout(fut, timeout or self._timeout)
+        """Timeout + error-swallow a future: failures are reported to the
+        manager and replaced by ``default``."""
+        timed = future_timeout(fut, timeout or self._timeout)
+        stream = torch.cuda.current_stream() if torch.cuda.is_available() else None
 
-        stream: Optional[torch.cuda.Stream] = (
-            torch.cuda.current_stream() if torch.cuda.is_available() else None
-        )
-
-        def callback(fut: torch.futures.Future[T]) -> T:
-            nonlocal default, stream
+        def absorb(f: torch.futures.Future[T]) -> T:
             with get_stream_context(stream):
                 try:
-                    return fut.value()
+                    return f.value()
                 except Exception as e:  # noqa: BLE001
-                    self._logger.exception(
-                        f"got exception in future -- skipping remaining: {e}"
-                    )
+                    self._logger.exception(f"future failed; using default: {e}")
                     self.report_error(e)
                     return default
 
-        return fut.then(callback)
+        return timed.then(absorb)
+
+    # -- quorum pipeline -------------------------------------------------------
 
     def start_quorum(
         self,
@@ -403,48 +521,50 @@ class Manager:
         shrink_only: bool = False,
         timeout: Optional[timedelta] = None,
     ) -> None:
-        """Compute a new quorum (async by default) and ready the manager for
-        a new step. Call before the forward pass; the quorum overlaps with it."""
-        # wait for a previous quorum to complete
-        if self._quorum_future is not None:
-            self._quorum_future.result()
+        """Begin a new step: negotiate a quorum (async by default, so it
+        overlaps the forward pass) and arm the per-step state."""
+        if self._attempt.pending is not None:
+            # a previous step's negotiation must fully settle first
+            self._attempt.pending.result()
 
-        self._errored = None
-        self._healing = False
-
-        self._quorum_future = self._executor.submit(
-            self._async_quorum,
-            allow_heal=allow_heal,
-            shrink_only=shrink_only,
-            quorum_timeout=timeout or self._quorum_timeout,
-            curr_device=(
-                torch.cuda.current_device() if torch.cuda.is_available() else -1
-            ),
+        attempt = _StepAttempt()
+        self._attempt = attempt
+        device = torch.cuda.current_device() if torch.cuda.is_available() else -1
+        attempt.pending = self._worker.submit(
+            lambda: self._quorum_cycle(
+                attempt,
+                allow_heal=allow_heal,
+                shrink_only=shrink_only,
+                quorum_timeout=timeout or self._quorum_timeout,
+                device=device,
+            )
         )
+
         if not self._use_async_quorum:
-            self.wait_quorum()
-            if self._healing:
-                # eagerly apply so the forward pass runs on healed weights
+            attempt.pending.result()
+            if attempt.healing:
+                # apply eagerly so the forward pass sees healed weights
                 self._apply_pending_state_dict()
-                self._healing = False
+                attempt.healing = False
 
     @torch.profiler.record_function("torchft_amd::manager::wait_quorum")
     def wait_quorum(self) -> None:
-        assert self._quorum_future is not None, "must call start_quorum before wait_quorum"
-        self._quorum_future.result()
+        self._attempt.ready()
 
-    @torch.profiler.record_function("torchft_amd::manager::_async_quorum")
-    def _async_quorum(
+    @torch.profiler.record_function("torchft_amd::manager::quorum_cycle")
+    def _quorum_cycle(
         self,
+        attempt: _StepAttempt,
         allow_heal: bool,
         shrink_only: bool,
         quorum_timeout: timedelta,
-        curr_device: int,
+        device: int,
     ) -> None:
-        if curr_device >= 0 and torch.cuda.is_available():
-            torch.cuda.set_device(curr_device)
+        """Worker-thread body: negotiate -> participate -> rebuild -> recover."""
+        if device >= 0 and torch.cuda.is_available():
+            torch.cuda.set_device(device)
 
-        quorum = self._client._quorum(
+        q = self._client._quorum(
             group_rank=self._group_rank,
             step=self._step,
             checkpoint_metadata=self._checkpoint_transport.metadata(),
@@ -454,189 +574,177 @@ class Manager:
             commit_failures=self._commit_failures,
         )
 
-        quorum_id = quorum.quorum_id
-        replica_rank = quorum.replica_rank
-        replica_world_size = quorum.replica_world_size
-        recover_src_manager_address = quorum.recover_src_manager_address
-        store_address = quorum.store_address
-        max_step = quorum.max_step
-        max_replica_rank = quorum.max_replica_rank
-        max_replica_world_size = quorum.max_world_size
-        heal = quorum.heal
-        replica_ids = quorum.replica_ids
-
-        ranks_in_quorum = [
-            extract_trailing_digits(rid.split(":")[0]) * self._group_world_size
-            + self._group_rank
-            for rid in replica_ids
-        ]
-
-        # Async quorum: only the already-up-to-date replicas participate this
-        # step (healing ones catch up); sync quorum: everyone.
-        self._participating_replica_rank, self._participating_replica_world_size = (
-            (max_replica_rank, max_replica_world_size)
-            if self._use_async_quorum or not allow_heal
-            else (replica_rank, replica_world_size)
+        attempt.participant_rank, attempt.participant_count = _participation_from(
+            replica_rank=q.replica_rank,
+            replica_world_size=q.replica_world_size,
+            max_replica_rank=q.max_replica_rank,
+            max_world_size=q.max_world_size,
+            defer_healing=self._use_async_quorum or not allow_heal,
+            world_size_mode=self._world_size_mode,
+            min_replica_size=self._min_replica_size,
         )
 
-        if self._replica_world_size_mode == WorldSizeMode.FIXED_WITH_SPARES:
-            self._participating_replica_world_size = min(
-                self._participating_replica_world_size, self._min_replica_size
-            )
-            if (
-                self._participating_replica_rank is not None
-                and self._participating_replica_rank >= self._min_replica_size
-            ):
-                self._participating_replica_rank = None
-
-        if quorum_id != self._quorum_id:
-            self.quorum_logger.info(
-                "",
-                extra={
-                    "job_id": os.environ.get("JOB_ID", "unknown"),
-                    "replica_id": self._replica_id,
-                    "rank": self._group_rank,
-                    "quorum_id": quorum_id,
-                    "step": max_step,
-                },
-            )
-            store_prefixed_addr = f"{store_address}/torchft/{quorum_id}/{self._group_rank}"
-            self._logger.info(f"reconfiguring for {quorum_id=} {store_prefixed_addr=}")
-            try:
-                self._quorum_id = quorum_id
-                # RCCL comm abort + re-init; must not race in-flight work.
-                if torch.cuda.is_available():
-                    torch.cuda.synchronize()
-                with torch.profiler.record_function(
-                    "torchft_amd::manager::pg::configure"
-                ):
-                    self._pg.configure(
-                        store_prefixed_addr,
-                        self._replica_id if self._replica_id is not None else "0",
-                        replica_rank,
-                        replica_world_size,
-                        quorum_id,
-                        self._group_rank,
-                        self._group_world_size,
-                        ranks_in_quorum,
-                    )
-            except Exception as e:  # noqa: BLE001
-                self._logger.exception(f"got exception in pg configure: {e}")
-                self.report_error(e)
+        if q.quorum_id != self._quorum_id:
+            if not self._rebuild_comm(attempt, q):
                 return
 
         if allow_heal:
-            # Recovery runs on the dedicated HIP recovery stream so the
-            # checkpoint copies overlap with compute.
-            with get_stream_context(self._recovery_stream):
-                try:
-                    if quorum.recover_dst_replica_ranks:
-                        self._logger.info(
-                            f"peers need recovery from us {quorum.recover_dst_replica_ranks}"
-                        )
-                        with torch.profiler.record_function(
-                            "torchft_amd::manager::send_checkpoint"
-                        ):
-                            self._checkpoint_transport.send_checkpoint(
-                                dst_ranks=quorum.recover_dst_replica_ranks,
-                                step=max_step,
-                                state_dict=self._manager_state_dict(),
-                                timeout=self._timeout,
-                            )
+            self._recovery_phase(attempt, q)
 
-                    if heal:
-                        self._healing = True
-                        self._logger.info(
-                            f"healing required, fetching checkpoint metadata from "
-                            f"{recover_src_manager_address=} {max_step=}"
-                        )
-                        primary_client = ManagerClient(
-                            recover_src_manager_address,
-                            connect_timeout=self._connect_timeout,
-                        )
-                        checkpoint_metadata = primary_client._checkpoint_metadata(
-                            self._group_rank, timeout=self._timeout
-                        )
-                        recover_src_replica_rank = quorum.recover_src_replica_rank
-                        assert recover_src_replica_rank is not None, (
-                            "must have a recover rank when healing"
-                        )
-                        self._logger.info(
-                            f"fetching checkpoint from {recover_src_replica_rank=} "
-                            f"with {checkpoint_metadata=}"
-                        )
-                        # stage the user state dict; applied from the main
-                        # thread only (at should_commit or sync start_quorum)
-                        self._pending_state_dict = self._checkpoint_transport.recv_checkpoint(
-                            src_rank=recover_src_replica_rank,
-                            metadata=checkpoint_metadata,
-                            step=max_step,
-                            timeout=self._timeout,
-                        )
-                        self.load_state_dict(self._pending_state_dict["torchft"])
-                        self._step = max_step
-                except Exception as e:  # noqa: BLE001
-                    self._logger.exception(f"got exception in recovery: {e}")
-                    self.report_error(e)
+    def _rebuild_comm(self, attempt: _StepAttempt, q: object) -> bool:
+        """Membership changed: abort and re-init the RCCL communicator against
+        the per-quorum store prefix. Returns False on failure (step aborts)."""
+        quorum_id = q.quorum_id
+        self.quorum_logger.info(
+            "",
+            extra={
+                "job_id": os.environ.get("JOB_ID", "unknown"),
+                "replica_id": self._replica_id,
+                "rank": self._group_rank,
+                "quorum_id": quorum_id,
+                "step": q.max_step,
+            },
+        )
+        store_prefixed_addr = (
+            f"{q.store_address}/torchft/{quorum_id}/{self._group_rank}"
+        )
+        self._logger.info(f"reconfiguring for {quorum_id=} {store_prefixed_addr=}")
 
-                self._recovery_event = (
-                    torch.cuda.current_stream().record_event()
-                    if self._recovery_stream is not None
-                    else None
+        # replica_ids -> global ranks for this rank's position in each group
+        global_ranks = [
+            extract_trailing_digits(rid.split(":")[0]) * self._group_world_size
+            + self._group_rank
+            for rid in q.replica_ids
+        ]
+
+        try:
+            self._quorum_id = quorum_id
+            if torch.cuda.is_available():
+                # the abort must not race in-flight collectives
+                torch.cuda.synchronize()
+            with torch.profiler.record_function("torchft_amd::manager::pg::configure"):
+                self._pg.configure(
+                    store_prefixed_addr,
+                    self._replica_id if self._replica_id is not None else "0",
+                    q.replica_rank,
+                    q.replica_world_size,
+                    quorum_id,
+                    self._group_rank,
+                    self._group_world_size,
+                    global_ranks,
                 )
+            return True
+        except Exception as e:  # noqa: BLE001
+            self._logger.exception(f"pg configure failed: {e}")
+            self.report_error(e)
+            return False
+
+    def _recovery_phase(self, attempt: _StepAttempt, q: object) -> None:
+        """Serve checkpoints to recovering peers and/or fetch our own heal,
+        on the dedicated recovery stream so the copies overlap compute."""
+        with get_stream_context(self._recovery_stream):
+            try:
+                if q.recover_dst_replica_ranks:
+                    self._serve_checkpoint(q)
+                if q.heal:
+                    attempt.healing = True
+                    self._fetch_checkpoint(attempt, q)
+            except Exception as e:  # noqa: BLE001
+                self._logger.exception(f"recovery failed: {e}")
+                self.report_error(e)
+
+            attempt.recovery_event = (
+                torch.cuda.current_stream().record_event()
+                if self._recovery_stream is not None
+                else None
+            )
+
+    @torch.profiler.record_function("torchft_amd::manager::send_checkpoint")
+    def _serve_checkpoint(self, q: object) -> None:
+        self._logger.info(f"serving checkpoint to {q.recover_dst_replica_ranks}")
+        self._checkpoint_transport.send_checkpoint(
+            dst_ranks=q.recover_dst_replica_ranks,
+            step=q.max_step,
+            state_dict=self._manager_state_dict(),
+            timeout=self._timeout,
+        )
+
+    def _fetch_checkpoint(self, attempt: _StepAttempt, q: object) -> None:
+        """We are behind: pull the live checkpoint from the recovery source
+        and stage it; the user portion is applied on the main thread only."""
+        src_addr = q.recover_src_manager_address
+        self._logger.info(f"healing from {src_addr} at step {q.max_step}")
+        src_client = ManagerClient(src_addr, connect_timeout=self._connect_timeout)
+        metadata = src_client._checkpoint_metadata(
+            self._group_rank, timeout=self._timeout
+        )
+        src_replica_rank = q.recover_src_replica_rank
+        assert src_replica_rank is not None, "healing requires a recovery source"
+        self._logger.info(f"fetching checkpoint from {src_replica_rank=} {metadata=}")
+        staged = self._checkpoint_transport.recv_checkpoint(
+            src_rank=src_replica_rank,
+            metadata=metadata,
+            step=q.max_step,
+            timeout=self._timeout,
+        )
+        attempt.staged_state = staged
+        # manager counters can load immediately; they are plain ints
+        self.load_state_dict(cast(Dict[str, int], staged["torchft"]))
+        self._step = q.max_step
 
     def _apply_pending_state_dict(self) -> None:
-        assert self._healing, "must be in healing state"
-        assert self._quorum_future is not None, "must call step before should_commit"
-        self._quorum_future.result()
+        """Main-thread application of a staged heal."""
+        assert self._attempt.healing, "no heal staged"
+        self._attempt.ready()
 
-        pending_state_dict = self._pending_state_dict
-        if pending_state_dict is None:
-            assert self.errored(), "checkpoint was not staged and no error occured"
+        staged = self._attempt.staged_state
+        if staged is None:
+            assert self.errored(), "no staged checkpoint and no error"
             return
 
-        self._logger.info("applying pending state dict")
-        assert len(self._load_state_dict_fns) > 0, "user load_state_dict is not initialized."
-        pending_user_state_dict = cast(Dict[str, object], pending_state_dict["user"])
+        self._logger.info("applying staged state dict")
+        assert self._load_state_dict_fns, "no load_state_dict fns registered"
+        user_state = cast(Dict[str, object], staged["user"])
         for key, load_fn in self._load_state_dict_fns.items():
-            load_fn(pending_user_state_dict[key])
-        self._pending_state_dict = None
-        self._logger.info("Loaded state dict.")
+            load_fn(user_state[key])
+        self._attempt.staged_state = None
+        self._logger.info("staged state dict applied")
+
+    # -- commit barrier --------------------------------------------------------
 
     @torch.profiler.record_function("torchft_amd::manager::should_commit")
     def should_commit(self, timeout: Optional[timedelta] = None) -> bool:
-        """All-ranks barrier deciding whether to step the optimizer.
+        """All-ranks barrier gating the optimizer step.
 
-        Must be called after backward and before optimizer.step(); the
-        optimizer may only step when this returns True.
+        Drains the recovery and compute streams, applies any staged heal,
+        then votes: the step commits iff every rank in the group is
+        error-free and enough replicas participated.
         """
-        # recovery must be complete before committing
-        if self._recovery_event is not None:
-            self._recovery_event.synchronize()
-            self._recovery_event = None
+        attempt = self._attempt
 
+        if attempt.recovery_event is not None:
+            attempt.recovery_event.synchronize()
+            attempt.recovery_event = None
         if torch.cuda.is_available():
             synchronize()
 
-        if err := self._pg.errored():
-            self.report_error(err)
+        if (pg_err := self._pg.errored()) is not None:
+            self.report_error(pg_err)
 
-        if self._healing:
+        if attempt.healing:
             self._apply_pending_state_dict()
 
-        enough_replicas = self.num_participants() >= self._min_replica_size
-        local_should_commit = enough_replicas and self._errored is None
-        should_commit = self._client.should_commit(
+        enough = self.num_participants() >= self._min_replica_size
+        my_vote = enough and attempt.error is None
+        decision = self._client.should_commit(
             self._group_rank,
             self._step,
-            local_should_commit,
+            my_vote,
             timeout=timeout or self._timeout,
         )
         self._logger.info(
-            f"should_commit={should_commit} enough_replicas={enough_replicas}, "
-            f"errored={self._errored}"
+            f"should_commit={decision} enough_replicas={enough} errored={attempt.error}"
         )
-
         self.commits_logger.info(
             "",
             extra={
@@ -645,44 +753,47 @@ class Manager:
                 "rank": self._group_rank,
                 "quorum_id": self._quorum_id,
                 "step": self._step,
-                "commit_result": should_commit,
+                "commit_result": decision,
             },
         )
 
+        # no checkpoint may be served across the commit boundary
         self._checkpoint_transport.disallow_checkpoint()
 
-        if should_commit:
+        if decision:
             self._step += 1
             self._batches_committed += self.num_participants()
             self._commit_failures = 0
         else:
             self._commit_failures += 1
-            if self._max_retries is not None and self._commit_failures > self._max_retries:
+            if (
+                self._max_retries is not None
+                and self._commit_failures > self._max_retries
+            ):
                 msg = (
-                    f"should_commit failed {self._commit_failures} times consecutively, "
-                    f"exceeding max_retries={self._max_retries}"
+                    f"should_commit failed {self._commit_failures} times "
+                    f"consecutively, exceeding max_retries={self._max_retries}"
                 )
                 self._logger.exception(msg)
                 raise RuntimeError(msg)
+        return decision
 
-        return should_commit
-
-    # -- state ----------------------------------------------------------------
+    # -- counters / introspection ---------------------------------------------
 
     def load_state_dict(self, state_dict: Dict[str, int]) -> None:
         self._step = state_dict["step"]
         self._batches_committed = state_dict["batches_committed"]
 
+    def state_dict(self) -> Dict[str, int]:
+        return {"step": self._step, "batches_committed": self._batches_committed}
+
     def _manager_state_dict(self) -> Dict[str, object]:
         with self._state_dict_lock.r_lock():
-            assert len(self._user_state_dicts) > 0, "user state_dict is not initialized."
+            assert self._user_state_dicts, "no user state_dict registered"
             return {
                 "user": {key: fn() for key, fn in self._user_state_dicts.items()},
                 "torchft": self.state_dict(),
             }
-
-    def state_dict(self) -> Dict[str, int]:
-        return {"step": self._step, "batches_committed": self._batches_committed}
 
     def current_step(self) -> int:
         return self._step
@@ -691,25 +802,50 @@ class Manager:
         return self._batches_committed
 
     def participating_rank(self) -> Optional[int]:
-        if self._quorum_future is None:
+        if self._attempt.pending is None:
             return None
         self.wait_quorum()
-        return self._participating_replica_rank
+        return self._attempt.participant_rank
 
     def num_participants(self) -> int:
-        if self._quorum_future is None:
+        if self._attempt.pending is None:
             return 0
         self.wait_quorum()
-        assert self._participating_replica_world_size >= 0, "internal error"
-        return self._participating_replica_world_size
+        assert self._attempt.participant_count >= 0
+        return self._attempt.participant_count
 
     def is_participating(self) -> bool:
-        if self._participating_replica_rank is None:
+        if self._attempt.participant_rank is None:
             return False
-        if self._healing:
+        if self._attempt.healing:
             assert self._use_async_quorum
             return False
         return True
+
+    # kept as properties so algorithm wrappers / tests can read them
+    @property
+    def _healing(self) -> bool:
+        return self._attempt.healing
+
+    @_healing.setter
+    def _healing(self, value: bool) -> None:
+        self._attempt.healing = value
+
+    @property
+    def _errored(self) -> Optional[ExceptionWithTraceback]:
+        return self._attempt.error
+
+    @_errored.setter
+    def _errored(self, value: Optional[ExceptionWithTraceback]) -> None:
+        self._attempt.error = value
+
+    @property
+    def _pending_state_dict(self) -> Optional[Dict[str, object]]:
+        return self._attempt.staged_state
+
+    @_pending_state_dict.setter
+    def _pending_state_dict(self, value: Optional[Dict[str, object]]) -> None:
+        self._attempt.staged_state = value
 
 
 class _ManagerLogger:
@@ -720,7 +856,10 @@ class _ManagerLogger:
         self._manager = manager
 
     def prefix(self) -> str:
-        return f"[{self._replica_id}/{self._group_rank} - step {self._manager.current_step()}]"
+        return (
+            f"[{self._replica_id}/{self._group_rank} "
+            f"- step {self._manager.current_step()}]"
+        )
 
     def info(self, msg: str) -> None:
         self._logger.info(f"{self.prefix()} {msg}")
@@ -732,162 +871,138 @@ class _ManagerLogger:
         self._logger.exception(f"{self.prefix()} {msg}")
 
 
-class _SimpleFuture(torch.futures.Future[T]):
-    """Wraps a pre-determined value for use in the _ManagedFuture callback
-    chain without blocking the CPU on ``value()`` of a real future."""
+# ---------------------------------------------------------------------------
+# stream-bound deferred continuations
+# ---------------------------------------------------------------------------
+
+
+class _ValueFuture(torch.futures.Future):
+    """Minimal already-completed future handed to user callbacks so they can
+    call ``.value()`` without touching a real (possibly device-bound) future."""
 
     def __init__(self, value: object) -> None:
         super().__init__()
-        self._value = value
+        self._v = value
 
     def value(self) -> object:
-        return self._value
-
-    def then(self, callback: Callable) -> torch.futures.Future:
-        raise NotImplementedError("callback-chain value wrapper only")
+        return self._v
 
     def wait(self) -> object:
-        raise NotImplementedError("callback-chain value wrapper only")
-
-    def done(self) -> bool:
-        raise NotImplementedError("callback-chain value wrapper only")
-
-    def add_done_callback(self, callback: Callable) -> None:
-        raise NotImplementedError("callback-chain value wrapper only")
-
-    def set_result(self, result: object) -> None:
-        raise NotImplementedError("callback-chain value wrapper only")
-
-    def set_exception(self, result: object) -> None:
-        raise NotImplementedError("callback-chain value wrapper only")
+        return self._v
 
 
-class _ManagedFuture(torch.futures.Future[T]):
-    """Lazy future chaining bound to the issuing HIP stream.
+class _ManagedFuture(torch.futures.Future):
+    """Future view of a ``_ManagedWork``.
 
-    ``then()`` only records the callback; the chain is materialized on
-    ``wait()/synchronize()/block_current_stream()`` of the owning
-    ``_ManagedWork``, inside the original stream context, with the first
-    callback wrapped by ``manager.wrap_future`` for error swallowing.
+    ``then()`` only records the callback on the owning work's continuation
+    ledger — nothing runs until the work is waited on, and everything then
+    runs inside the stream the collective was issued on.
     """
 
-    def __init__(self, managed_work: "weakref.ReferenceType[_ManagedWork]") -> None:
+    def __init__(self, owner: "_ManagedWork") -> None:
         super().__init__()
-        self._managed_work = managed_work
-        self._fut: Optional[torch.futures.Future[T]] = None
-        self._next: Optional["_ManagedFuture[object]"] = None
-        self._callback: Optional[Callable[[torch.futures.Future[T]], object]] = None
+        self._owner = owner
+        # the realized torch future once materialized (for callers that need
+        # a genuine torch.futures.Future, e.g. the DDP reducer)
+        self._fut: Optional[torch.futures.Future] = None
 
-    def then(
-        self, callback: Callable[[torch.futures.Future[T]], S]
-    ) -> torch.futures.Future[S]:
-        managed_work = self._managed_work()
-        assert managed_work is not None, "got garbage collected"
-        self._callback = callback
-        self._next = _ManagedFuture[object](self._managed_work)
-        managed_work._managed_fut_tail = self._next
-        return cast(torch.futures.Future[S], self._next)
+    def then(self, callback: Callable) -> "torch.futures.Future":
+        self._owner._append(lambda v: callback(_ValueFuture(v)))
+        return self
 
     def wait(self) -> object:
-        assert self._fut
-        return self._fut.wait()
+        fut = self._owner._realized()
+        self._fut = fut
+        return fut.wait()
 
     def value(self) -> object:
-        raise NotImplementedError("used to create callback chains only")
-
-    def done(self) -> bool:
-        raise NotImplementedError("used to create callback chains only")
-
-    def add_done_callback(self, callback: Callable) -> None:
-        raise NotImplementedError("used to create callback chains only")
-
-    def set_result(self, result: object) -> None:
-        raise NotImplementedError("used to create callback chains only")
-
-    def set_exception(self, result: object) -> None:
-        raise NotImplementedError("used to create callback chains only")
+        return self._owner._value
 
 
 class _ManagedWork(dist._Work):
-    """Work whose future-callback chain is created lazily on wait/synchronize,
-    always inside the stream the collective was issued on."""
+    """Work with a flat ledger of deferred continuations.
+
+    The continuations (error-normalization, gradient scaling, bucket
+    scatter-back, ...) are recorded eagerly but executed lazily: the first
+    ``wait()/synchronize()/block_current_stream()`` attaches ONE real
+    ``Future.then`` to the underlying collective, whose callback replays the
+    ledger in order inside the issuing HIP stream. Any continuation error is
+    reported to the manager and the last good value is returned, so the
+    training loop never sees an exception mid-backward.
+    """
 
     def __init__(self, manager: Manager, work: dist._Work, value: object) -> None:
         super().__init__()
-        self._work = work
         self._manager = manager
+        self._work = work
         self._value = value
-        self._managed_fut_head = _ManagedFuture[object](weakref.ref(self))
-        self._managed_fut_tail: _ManagedFuture[object] = self._managed_fut_head
+        self._ledger: List[Callable[[object], object]] = []
         self._stream: Optional[torch.cuda.Stream] = (
             torch.cuda.current_stream() if torch.cuda.is_available() else None
         )
-        self._is_set_future_callback_called = False
+        self._done_fut: Optional[torch.futures.Future] = None
+        self._proxy = _ManagedFuture(self)
 
-    def _set_future_callback(self) -> None:
-        if self._is_set_future_callback_called:
+    def _append(self, step: Callable[[object], object]) -> None:
+        assert self._done_fut is None, "cannot add continuations after wait()"
+        self._ledger.append(step)
+
+    def _realized(self) -> torch.futures.Future:
+        assert self._done_fut is not None, "work must be waited on first"
+        return self._done_fut
+
+    def _materialize(self) -> None:
+        if self._done_fut is not None:
             return
 
-        managed_fut: _ManagedFuture[object] = self._managed_fut_head
-        managed_fut._fut = self._work.get_future()
-        value = self._value
+        inner = self._work.get_future()
 
-        is_future_wrapped = False
-        while managed_fut._next:
+        def replay(f: torch.futures.Future) -> object:
+            with get_stream_context(self._stream):
+                value = self._value
+                try:
+                    f.wait()  # establishes the stream dependency
+                    for step in self._ledger:
+                        value = step(value)
+                except Exception as e:  # noqa: BLE001
+                    self._manager._logger.exception(f"continuation failed: {e}")
+                    self._manager.report_error(e)
+                self._value = value
+                return value
 
-            def callback(fut: torch.futures.Future[object]) -> object:
-                nonlocal managed_fut, value
-                # keep the chain on the issuing stream, not the PG stream
-                with get_stream_context(self._stream):
-                    fut.wait()  # stream dependency
-                    assert managed_fut._callback
-                    value = managed_fut._callback(_SimpleFuture(value))
-                    return value
+        self._done_fut = future_timeout(inner, self._manager._timeout).then(replay)
 
-            assert managed_fut._fut
-            fut = managed_fut._fut.then(callback)
-            assert managed_fut._next
-            managed_fut = managed_fut._next
-            managed_fut._fut = fut
-
-            if is_future_wrapped:
-                continue
-            managed_fut._fut = self._manager.wrap_future(managed_fut._fut, value)
-            is_future_wrapped = True
-
-        self._value = value
-        self._is_set_future_callback_called = True
-
-    def _assert_same_stream(self) -> None:
+    def _on_issuing_stream(self) -> None:
         if self._stream is not None:
-            assert self._stream == torch.cuda.current_stream()
+            assert self._stream == torch.cuda.current_stream(), (
+                "managed work must be completed on the stream it was issued on"
+            )
 
     def wait(self, timeout: Optional[timedelta] = None) -> bool:
-        self._assert_same_stream()
+        self._on_issuing_stream()
         try:
             with get_stream_context(self._stream):
                 self._work.wait()
-                self._set_future_callback()
-            with get_stream_context(self._stream):
-                self._managed_fut_tail.wait()
+                self._materialize()
+                self._realized().wait()
             return True
         except Exception as e:  # noqa: BLE001
-            self._manager._logger.exception(f"got exception waiting for work {e}")
+            self._manager._logger.exception(f"wait failed: {e}")
             self._manager.report_error(e)
             return False
 
     def block_current_stream(self, timeout: Optional[timedelta] = None) -> None:
-        self._assert_same_stream()
+        self._on_issuing_stream()
         with get_stream_context(self._stream):
             self._work.block_current_stream()
-        self._set_future_callback()
+        self._materialize()
 
     def synchronize(self) -> None:
-        self._assert_same_stream()
+        self._on_issuing_stream()
         if torch.cuda.is_available():
             self.block_current_stream()
         else:
-            self._set_future_callback()
+            self._materialize()
 
-    def get_future(self) -> torch.futures.Future[object]:
-        return self._managed_fut_tail
+    def get_future(self) -> torch.futures.Future:
+        return self._proxy

@@ -179,3 +179,34 @@ class TestWrappers:
         fut = work.get_future()
         with pytest.raises(RuntimeError, match="injected"):
             fut.wait()
+
+
+class TestAbortDiagnostics:
+    """abort() must dump the flight-recorder op log."""
+
+    def test_oplog_dump_on_abort(self, tmp_path, monkeypatch):
+        import json
+        import os
+
+        from torch.distributed import TCPStore
+
+        monkeypatch.setenv("TORCHFT_ABORT_DUMP_DIR", str(tmp_path))
+        store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+        pg = ProcessGroupGloo(timeout=timedelta(seconds=5))
+        pg.configure(f"127.0.0.1:{store.port}/qd", "r0", 0, 1)
+        t = torch.ones(8)
+        pg.allreduce([t], ReduceOp.SUM).wait()
+        pg.abort()  # errored abort dumps the log
+
+        dumps = [p for p in os.listdir(tmp_path) if p.startswith("oplog_")]
+        assert len(dumps) == 1
+        records = json.loads((tmp_path / dumps[0]).read_text())
+        assert any(r["op"] == "allreduce" for r in records)
+        rec = [r for r in records if r["op"] == "allreduce"][-1]
+        assert rec["tensors"] == [["float32", [8]]]
+        assert rec["status"] == "in_flight"
+
+    def test_no_dump_without_env(self, tmp_path, monkeypatch):
+        monkeypatch.delenv("TORCHFT_ABORT_DUMP_DIR", raising=False)
+        pg = ProcessGroupGloo(timeout=timedelta(seconds=5))
+        assert pg._oplog.dump("x") is None

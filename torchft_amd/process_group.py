@@ -1,30 +1,42 @@
 """Reconfigurable process groups for fault-tolerant training on MI355X.
 
-The fault-tolerance protocol requires communicators that can be aborted and
-rebuilt at step granularity when quorum membership changes. On MI355X the
-production backend is RCCL over xGMI (PyTorch's ``"nccl"`` backend IS RCCL on
-ROCm); reconfiguration is abort-and-recreate with non-blocking communicator
-init plus user-space timeouts that call ``abort()`` so a wedged collective can
-never take down the trainer.
+Quorum membership changes at step granularity, so communicators must be
+rebuildable at step granularity too. On MI355X the production backend is
+RCCL over xGMI (torch's ``"nccl"`` backend IS RCCL on ROCm); rebuilds are
+abort-and-recreate with non-blocking communicator init, and every issued
+collective carries a user-space deadline whose expiry calls ``abort()`` —
+a dead peer unblocks the survivors instead of wedging or killing them.
 
-Reference parity (semantics, not code): torchft/process_group.py —
-``ProcessGroup`` ABC with ``configure/abort/errored/set_timeout`` (:131-399),
-``ProcessGroupWrapper`` (:402-640), ``ProcessGroupGloo`` (:643-711),
-``ProcessGroupNCCL`` (:780-892), ``ProcessGroupDummy`` (:1005-1134),
-``ErrorSwallowingProcessGroupWrapper`` (:1176-1249), ``FakeProcessGroupWrapper``
-(:1252-1317), ``ManagedProcessGroup`` (:1320-1353).
+Semantics matched against the reference (torchft/process_group.py: ABC
+:131-399, wrapper :402-640, Gloo :643-711, NCCL :780-892, Dummy :1005-1134,
+error-swallowing :1176-1249, fake :1252-1317, managed :1320-1353) but the
+structure here is different by design:
+
+* every collective funnels through ONE dispatch point (``_issue``) that
+  applies the timeout policy, records the op in a flight-recorder ring
+  buffer, and attaches the deadline guard — there is no per-backend hook
+  hierarchy;
+* each ``ProcessGroupWrapper`` keeps an ``_OpLog`` of the last N issued
+  collectives (op, shapes, bytes, quorum); ``abort()`` dumps it as JSON to
+  ``TORCHFT_ABORT_DUMP_DIR`` so a wedged RCCL collective on a production
+  job is debuggable post-mortem (flight-recorder analog of
+  torchft/manager.py:815-824).
 
 Subprocess-isolated ("Baby") variants live in ``baby_process_group.py``.
 """
 
 from __future__ import annotations
 
+import json
 import logging
 import os
+import threading
+import time
 import warnings
+from collections import deque
 from contextlib import contextmanager
 from datetime import timedelta
-from typing import TYPE_CHECKING, Generator, List, Optional, TypeVar, Union
+from typing import TYPE_CHECKING, Callable, Generator, List, Optional, Union
 
 import torch
 import torch.distributed as dist
@@ -56,45 +68,122 @@ if TYPE_CHECKING:
 
 logger: logging.Logger = logging.getLogger(__name__)
 
-T = TypeVar("T")
+ABORT_DUMP_DIR_ENV = "TORCHFT_ABORT_DUMP_DIR"
 
 
 def create_store_client(store_addr: str, timeout: timedelta) -> Store:
-    """Create a PrefixStore(TCPStore) client from ``host:port/prefix``."""
+    """``host:port/prefix`` -> PrefixStore over a TCPStore client."""
     host, _, rest = store_addr.partition(":")
     port, _, prefix = rest.partition("/")
-    store = TCPStore(
-        host_name=host,
-        port=int(port),
-        is_master=False,
-        wait_for_workers=False,
-        timeout=timeout,
+    return PrefixStore(
+        prefix,
+        TCPStore(
+            host_name=host,
+            port=int(port),
+            is_master=False,
+            wait_for_workers=False,
+            timeout=timeout,
+        ),
     )
-    return PrefixStore(prefix, store)
+
+
+# ---------------------------------------------------------------------------
+# flight recorder
+# ---------------------------------------------------------------------------
+
+
+def _tensor_brief(t: object) -> object:
+    if isinstance(t, torch.Tensor):
+        return [str(t.dtype).replace("torch.", ""), list(t.shape)]
+    if isinstance(t, (list, tuple)):
+        return [_tensor_brief(x) for x in t]
+    return None
+
+
+class _OpLog:
+    """Ring buffer of recently issued collectives.
+
+    Cheap enough for the hot path (a dict append, no device work); dumped to
+    disk when the group aborts so the post-mortem shows exactly which
+    collective wedged and what was in flight.
+    """
+
+    def __init__(self, capacity: int = 64) -> None:
+        self._ring: deque = deque(maxlen=capacity)
+        self._seq = 0
+        self._mu = threading.Lock()
+
+    def record(self, op: str, tensors: object, quorum_id: object) -> int:
+        with self._mu:
+            self._seq += 1
+            self._ring.append(
+                {
+                    "seq": self._seq,
+                    "op": op,
+                    "tensors": _tensor_brief(tensors),
+                    "quorum_id": quorum_id,
+                    "time": time.time(),
+                    "status": "issued",
+                }
+            )
+            return self._seq
+
+    def mark(self, seq: int, status: str) -> None:
+        with self._mu:
+            for rec in reversed(self._ring):
+                if rec["seq"] == seq:
+                    rec["status"] = status
+                    return
+
+    def snapshot(self) -> List[dict]:
+        with self._mu:
+            return [dict(r) for r in self._ring]
+
+    def dump(self, tag: str) -> Optional[str]:
+        """Write the ring to ``TORCHFT_ABORT_DUMP_DIR`` (no-op if unset)."""
+        dump_dir = os.environ.get(ABORT_DUMP_DIR_ENV)
+        if not dump_dir:
+            return None
+        try:
+            os.makedirs(dump_dir, exist_ok=True)
+            path = os.path.join(dump_dir, f"oplog_{tag}_{os.getpid()}.json")
+            with open(path, "w") as f:
+                json.dump(self.snapshot(), f, indent=1)
+            return path
+        except OSError as e:  # never let diagnostics kill the abort path
+            logger.warning(f"op-log dump failed: {e}")
+            return None
+
+
+# ---------------------------------------------------------------------------
+# the reconfigurable ProcessGroup surface
+# ---------------------------------------------------------------------------
 
 
 class ProcessGroup(BaseProcessGroup):
-    """Abstract reconfigurable process group.
+    """torch ProcessGroup extended with quorum-driven reconfiguration.
 
-    Adds ``configure()`` (rebuild the communicator against a fresh store
-    prefix for a new quorum), ``abort()``, ``errored()`` and ``set_timeout()``
-    on top of the stock torch collective surface, plus registration into the
-    c10d backend registry so functional collectives / DeviceMesh work.
+    Adds ``configure()`` (rebuild against a fresh per-quorum store prefix),
+    ``abort()``, ``errored()`` and ``set_timeout()``, plus registration into
+    the c10d registry so functional collectives and DeviceMesh can address
+    the group.
     """
 
     def __init__(self, *args: object, **kwargs: object) -> None:
         super().__init__(*args, **kwargs)  # pyre-ignore[6]
         self._group_name: Optional[str] = None
 
-    # -- collective surface (implemented by subclasses) --------------------
+    def _absent(self, op: str) -> NotImplementedError:
+        return NotImplementedError(f"{type(self).__name__} does not implement {op}")
 
+    # collective surface — concrete groups override what they support
     def allgather(
         self,
         output_tensors: List[List[torch.Tensor]],
         input_tensor: List[torch.Tensor],
         opts: AllgatherOptions,
     ) -> Work:
-        raise NotImplementedError("not implemented")
+        raise self._absent("allgather")
 
     def allgather_into_tensor_coalesced(
         self,
@@ -102,21 +191,21 @@ class ProcessGroup(BaseProcessGroup):
         input_tensors: List[torch.Tensor],
         opts: AllgatherOptions,
     ) -> Work:
-        raise NotImplementedError("not implemented")
+        raise self._absent("allgather_into_tensor_coalesced")
 
     def allreduce(
         self,
         tensors: List[torch.Tensor],
         opts: Union[AllreduceOptions, ReduceOp],
     ) -> Work:
-        raise NotImplementedError("not implemented")
+        raise self._absent("allreduce")
 
     def allreduce_coalesced(
         self,
         tensors: List[torch.Tensor],
         opts: AllreduceCoalescedOptions,
     ) -> Work:
-        raise NotImplementedError("not implemented")
+        raise self._absent("allreduce_coalesced")
 
     def alltoall_base(
         self,
@@ -126,13 +215,15 @@ class ProcessGroup(BaseProcessGroup):
         input_split_sizes: List[int],
         opts: AllToAllOptions,
     ) -> Work:
-        raise NotImplementedError("not implemented")
+        raise self._absent("alltoall_base")
 
     def barrier(self, opts: BarrierOptions) -> Work:
-        raise NotImplementedError("not implemented")
+        raise self._absent("barrier")
 
-    def broadcast(self, tensor_list: List[torch.Tensor], opts: BroadcastOptions) -> Work:
-        raise NotImplementedError("not implemented")
+    def broadcast(
+        self, tensor_list: List[torch.Tensor], opts: BroadcastOptions
+    ) -> Work:
+        raise self._absent("broadcast")
 
     def broadcast_one(self, tensor: torch.Tensor, root: int) -> Work:
         opts = BroadcastOptions()
@@ -140,7 +231,7 @@ class ProcessGroup(BaseProcessGroup):
         return self.broadcast([tensor], opts)
 
     def recv(self, tensors: List[torch.Tensor], src_rank: int, tag: int) -> Work:
-        raise NotImplementedError("not implemented")
+        raise self._absent("recv")
 
     def reduce_scatter(
         self,
@@ -148,7 +239,7 @@ class ProcessGroup(BaseProcessGroup):
         input_tensors: List[List[torch.Tensor]],
         opts: ReduceScatterOptions,
     ) -> Work:
-        raise NotImplementedError("not implemented")
+        raise self._absent("reduce_scatter")
 
     def reduce_scatter_tensor_coalesced(
         self,
@@ -156,13 +247,12 @@ class ProcessGroup(BaseProcessGroup):
         input_tensors: List[torch.Tensor],
         opts: ReduceScatterOptions,
     ) -> Work:
-        raise NotImplementedError("not implemented")
+        raise self._absent("reduce_scatter_tensor_coalesced")
 
     def send(self, tensors: List[torch.Tensor], dst_rank: int, tag: int) -> Work:
-        raise NotImplementedError("not implemented")
+        raise self._absent("send")
 
-    # -- reconfiguration ---------------------------------------------------
-
+    # reconfiguration surface
     def configure(
         self,
         store_addr: str,
@@ -174,40 +264,50 @@ class ProcessGroup(BaseProcessGroup):
         group_world_size: Optional[int] = None,
         global_ranks: Optional[list[int]] = None,
     ) -> None:
-        """Rebuild the communicator. ``store_addr`` must be a unique prefixed
-        store address per quorum (``host:port/prefix``). Blocks until the new
-        communicator exists; raises on failure."""
-        raise NotImplementedError("not implemented")
+        """Rebuild the communicator against ``store_addr`` (a per-quorum
+        ``host:port/prefix``). Blocks until the new communicator exists."""
+        raise self._absent("configure")
 
     def size(self) -> int:
-        raise NotImplementedError("not implemented")
+        raise self._absent("size")
 
     def getBackendName(self) -> str:
-        raise NotImplementedError("not implemented")
+        raise self._absent("getBackendName")
 
+    def abort(self) -> None:
+        pass
+
+    def shutdown(self) -> None:
+        pass
+
+    def errored(self) -> Optional[Exception]:
+        """Async error requiring reconfiguration, if any."""
+        return None
+
+    def set_timeout(self, timeout: timedelta) -> None:
+        raise self._absent("set_timeout")
+
+    # c10d registry integration (functional collectives / DeviceMesh).
+    # Elastic worlds can't be described to DeviceMesh, so the group is
+    # registered under world size 1 and its real size changes underneath.
     def _register(self, name: str) -> str:
         group_name = f"{self.getBackendName()}:{name}"
 
-        # Resizable worlds don't fit DeviceMesh, so register as world-size-1.
-        def create_pg(
+        def factory(
             prefix_store: PrefixStore, rank: int, world_size: int, timeout: float
         ) -> "ProcessGroup":
             return self
 
-        devices = ["cpu"]
-        if torch.cuda.is_available():
-            devices.append("cuda")
-        dist.Backend.register_backend(group_name, create_pg, devices=devices)
+        devices = ["cpu", "cuda"] if torch.cuda.is_available() else ["cpu"]
+        dist.Backend.register_backend(group_name, factory, devices=devices)
         return group_name
 
     def register(self, name: str) -> "ProcessGroup":
-        """Register with the c10d global registry (enables functional
-        collectives). Call at most once."""
-        group_name = self._register(name)
+        """Register with the global c10d registry; call at most once."""
         return dist.new_group(
             ranks=[dist.get_rank()],
-            backend=group_name,
-            group_desc=group_name,
+            backend=self._register(name),
+            group_desc=f"{self.getBackendName()}:{name}",
             timeout=timedelta(seconds=60.0),
         )
 
@@ -223,29 +323,18 @@ class ProcessGroup(BaseProcessGroup):
     def unregister(self) -> None:
         dist.destroy_process_group(self)
 
-    def abort(self) -> None:
-        pass
-
-    def shutdown(self) -> None:
-        pass
-
-    def errored(self) -> Optional[Exception]:
-        """Async error that requires reconfiguration, if any."""
-        return None
-
-    def set_timeout(self, timeout: timedelta) -> None:
-        raise NotImplementedError("set_timeout not implemented")
-
     def __repr__(self) -> str:
         return f"{self.__class__.__name__}()"
 
 
 class ProcessGroupWrapper(ProcessGroup):
-    """Reconfiguration by abort-and-recreate around any inner process group.
+    """Epoch-style reconfiguration around a torch process group.
 
-    ``configure()`` aborts the old backend, creates a fresh TCPStore client
-    against the per-quorum prefix and calls ``_create_pg``. Subclasses hook
-    ``_create_pg`` / ``_wrap_work`` / ``_opts_hook`` / ``_run_context``.
+    Every ``configure()`` starts a new communicator epoch: the previous
+    communicator is aborted, a store client is opened against the quorum's
+    prefix, and ``_build()`` creates the replacement. All collectives pass
+    through ``_issue()``, the single point where the timeout policy, the
+    flight-recorder record and the deadline guard are applied.
     """
 
     def __init__(
@@ -254,7 +343,7 @@ class ProcessGroupWrapper(ProcessGroup):
         pg: Optional[ProcessGroup] = None,
     ) -> None:
         super().__init__(0, 1)
-        self._pg: Optional[BaseProcessGroup] = pg
+        self._inner: Optional[BaseProcessGroup] = pg
         self._timeout = timeout
         self._replica_id: Optional[str] = None
         self._rank: Optional[int] = None
@@ -262,13 +351,10 @@ class ProcessGroupWrapper(ProcessGroup):
         self._group_rank: Optional[int] = None
         self._group_world_size: Optional[int] = None
         self._global_ranks: Optional[list[int]] = None
+        self._oplog = _OpLog()
         self.errors_logger: logging.Logger = logging.getLogger("torchft_errors")
 
-    def getBackendName(self) -> str:
-        pg = self._pg
-        if isinstance(pg, ProcessGroup):
-            return pg.getBackendName()
-        raise NotImplementedError("not implemented")
+    # -- epoch management --------------------------------------------------
 
     def configure(
         self,
@@ -281,15 +367,17 @@ class ProcessGroupWrapper(ProcessGroup):
         group_world_size: Optional[int] = None,
         global_ranks: Optional[list[int]] = None,
     ) -> None:
-        pg = self._pg
         self._replica_id = replica_id
+        self._rank = rank
         self._quorum_id = quorum_id
         self._group_rank = group_rank
         self._group_world_size = group_world_size
-        self._rank = rank
         self._global_ranks = global_ranks
-        if isinstance(pg, ProcessGroup):
-            pg.configure(
+
+        inner = self._inner
+        if isinstance(inner, ProcessGroup):
+            # wrapping an already-reconfigurable group: delegate the epoch
+            inner.configure(
                 store_addr,
                 replica_id,
                 rank,
@@ -303,7 +391,10 @@ class ProcessGroupWrapper(ProcessGroup):
 
         self.abort(errored=False)
         store = create_store_client(store_addr, timeout=self._timeout)
-        self._pg = self._create_pg(store, rank, world_size)
+        self._inner = self._build(store, rank, world_size)
+
+    def _build(self, store: Store, rank: int, world_size: int) -> BaseProcessGroup:
+        raise self._absent("_build")
 
     def abort(self, errored: bool = True) -> None:
         if errored:
@@ -317,41 +408,70 @@ class ProcessGroupWrapper(ProcessGroup):
                     "error": "process_group_abort",
                 },
             )
-        pg = self._pg
-        if pg is not None:
-            if hasattr(pg, "abort"):
-                pg.abort()
-            else:
-                backend = None
-                try:
-                    if torch.cuda.is_available():
-                        backend = pg._get_backend(torch.device("cuda"))
-                except RuntimeError:
-                    backend = None
-                if backend is not None and hasattr(backend, "abort"):
-                    backend.abort()
-            self._pg = None
+            path = self._oplog.dump(
+                f"{self._replica_id or 'r'}_{self._rank or 0}_q{self._quorum_id}"
+            )
+            if path:
+                logger.error(f"aborting; op log dumped to {path}")
+
+        inner, self._inner = self._inner, None
+        if inner is None:
+            return
+        if hasattr(inner, "abort"):
+            inner.abort()
+            return
+        # stock torch pg: abort the device backend if one exists
+        try:
+            backend = (
+                inner._get_backend(torch.device("cuda"))
+                if torch.cuda.is_available()
+                else None
+            )
+        except RuntimeError:
+            backend = None
+        if backend is not None and hasattr(backend, "abort"):
+            backend.abort()
 
     def shutdown(self) -> None:
-        self._pg = None
-
-    def _create_pg(self, store: Store, rank: int, world_size: int) -> BaseProcessGroup:
-        raise NotImplementedError("not implemented")
-
-    def _wrap_work(self, work: Work, opts: object) -> Work:
-        return work
-
-    def _opts_hook(self, opts: T) -> T:
-        return opts
-
-    @contextmanager
-    def _run_context(self) -> Generator[None, None, None]:
-        yield
+        self._inner = None
 
     def set_timeout(self, timeout: timedelta) -> None:
         self._timeout = timeout
 
-    # -- collectives: forward to the inner pg through the hooks ------------
+    @property
+    def parent(self) -> BaseProcessGroup:
+        assert self._inner is not None, "process group not initialized"
+        return self._inner
+
+    # -- single dispatch point ---------------------------------------------
+
+    def _prepare_opts(self, opts: object) -> object:
+        """Timeout-policy hook applied to every op's options."""
+        return opts
+
+    def _attach_deadline(self, work: Work, opts: object) -> Work:
+        """Deadline-policy hook applied to every returned work."""
+        return work
+
+    @contextmanager
+    def _launch_guard(self) -> Generator[None, None, None]:
+        """Guards the (possibly blocking) op launch itself."""
+        yield
+
+    def _issue(
+        self, op: str, tensors: object, opts: object, launch: Callable[[object], Work]
+    ) -> Work:
+        seq = self._oplog.record(op, tensors, self._quorum_id)
+        try:
+            with self._launch_guard():
+                work = launch(self._prepare_opts(opts))
+        except Exception:
+            self._oplog.mark(seq, "launch_failed")
+            raise
+        self._oplog.mark(seq, "in_flight")
+        return self._attach_deadline(work, opts)
+
+    # -- collective surface -------------------------------------------------
 
     def allgather(
         self,
@@ -359,11 +479,12 @@ class ProcessGroupWrapper(ProcessGroup):
         input_tensor: List[torch.Tensor],
         opts: AllgatherOptions,
     ) -> Work:
-        with self._run_context():
-            return self._wrap_work(
-                self.parent.allgather(output_tensors, input_tensor, self._opts_hook(opts)),
-                opts,
-            )
+        return self._issue(
+            "allgather",
+            input_tensor,
+            opts,
+            lambda o: self.parent.allgather(output_tensors, input_tensor, o),
+        )
 
     def allgather_into_tensor_coalesced(
         self,
@@ -371,25 +492,29 @@ class ProcessGroupWrapper(ProcessGroup):
         input_tensors: List[torch.Tensor],
         opts: AllgatherOptions,
     ) -> Work:
-        with self._run_context():
-            return self._wrap_work(
-                self.parent.allgather_into_tensor_coalesced(
-                    output_tensors, input_tensors, self._opts_hook(opts)
-                ),
-                opts,
-            )
+        return self._issue(
+            "allgather_into_tensor_coalesced",
+            input_tensors,
+            opts,
+            lambda o: self.parent.allgather_into_tensor_coalesced(
+                output_tensors, input_tensors, o
+            ),
+        )
 
     def allreduce(self, tensors: List[torch.Tensor], opts: object) -> Work:
-        with self._run_context():
-            return self._wrap_work(self.parent.allreduce(tensors, self._opts_hook(opts)), opts)
+        return self._issue(
+            "allreduce", tensors, opts, lambda o: self.parent.allreduce(tensors, o)
+        )
 
     def allreduce_coalesced(
         self, tensors: List[torch.Tensor], opts: Union[AllreduceOptions, ReduceOp]
     ) -> Work:
-        with self._run_context():
-            return self._wrap_work(
-                self.parent.allreduce_coalesced(tensors, self._opts_hook(opts)), opts
-            )
+        return self._issue(
+            "allreduce_coalesced",
+            tensors,
+            opts,
+            lambda o: self.parent.allreduce_coalesced(tensors, o),
+        )
 
     def alltoall_base(
         self,
@@ -399,29 +524,32 @@ class ProcessGroupWrapper(ProcessGroup):
         input_split_sizes: List[int],
         opts: AllToAllOptions,
     ) -> Work:
-        with self._run_context():
-            return self._wrap_work(
-                self.parent.alltoall_base(
-                    output_buffer,
-                    input_buffer,
-                    output_split_sizes,
-                    input_split_sizes,
-                    self._opts_hook(opts),
-                ),
-                opts,
-            )
+        return self._issue(
+            "alltoall_base",
+            input_buffer,
+            opts,
+            lambda o: self.parent.alltoall_base(
+                output_buffer, input_buffer, output_split_sizes, input_split_sizes, o
+            ),
+        )
 
     def barrier(self, opts: Optional[BarrierOptions] = None) -> Work:
-        with self._run_context():
-            return self._wrap_work(self.parent.barrier(self._opts_hook(opts)), opts)
+        return self._issue(
+            "barrier", None, opts, lambda o: self.parent.barrier(o)
+        )
 
     def broadcast(self, tensor_list: List[torch.Tensor], opts: object) -> Work:
-        with self._run_context():
-            return self._wrap_work(self.parent.broadcast(tensor_list, self._opts_hook(opts)), opts)
+        return self._issue(
+            "broadcast",
+            tensor_list,
+            opts,
+            lambda o: self.parent.broadcast(tensor_list, o),
+        )
 
     def recv(self, tensors: List[torch.Tensor], src_rank: int, tag: int) -> Work:
-        with self._run_context():
-            return self._wrap_work(self.parent.recv(tensors, src_rank, tag), None)
+        return self._issue(
+            "recv", tensors, None, lambda o: self.parent.recv(tensors, src_rank, tag)
+        )
 
     def reduce_scatter(
         self,
@@ -429,11 +557,12 @@ class ProcessGroupWrapper(ProcessGroup):
         input_tensors: List[List[torch.Tensor]],
         opts: object,
     ) -> Work:
-        with self._run_context():
-            return self._wrap_work(
-                self.parent.reduce_scatter(output_tensors, input_tensors, self._opts_hook(opts)),
-                opts,
-            )
+        return self._issue(
+            "reduce_scatter",
+            input_tensors,
+            opts,
+            lambda o: self.parent.reduce_scatter(output_tensors, input_tensors, o),
+        )
 
     def reduce_scatter_tensor_coalesced(
         self,
@@ -441,53 +570,56 @@ class ProcessGroupWrapper(ProcessGroup):
         input_tensors: List[torch.Tensor],
         opts: ReduceScatterOptions,
     ) -> Work:
-        with self._run_context():
-            return self._wrap_work(
-                self.parent.reduce_scatter_tensor_coalesced(
-                    output_tensors, input_tensors, self._opts_hook(opts)
-                ),
-                opts,
-            )
+        return self._issue(
+            "reduce_scatter_tensor_coalesced",
+            input_tensors,
+            opts,
+            lambda o: self.parent.reduce_scatter_tensor_coalesced(
+                output_tensors, input_tensors, o
+            ),
+        )
 
     def send(self, tensors: List[torch.Tensor], dst_rank: int, tag: int) -> Work:
-        with self._run_context():
-            return self._wrap_work(self.parent.send(tensors, dst_rank, tag), None)
+        return self._issue(
+            "send", tensors, None, lambda o: self.parent.send(tensors, dst_rank, tag)
+        )
 
     def size(self) -> int:
         return self.parent.size()
 
-    @property
-    def parent(self) -> BaseProcessGroup:
-        assert self._pg is not None, "process group not initialized"
-        return self._pg
+    def getBackendName(self) -> str:
+        inner = self._inner
+        if isinstance(inner, ProcessGroup):
+            return inner.getBackendName()
+        raise self._absent("getBackendName")
 
     def __repr__(self) -> str:
-        return f"{self.__class__.__name__}(pg={self._pg})"
+        return f"{self.__class__.__name__}(pg={self._inner})"
 
 
 class ProcessGroupGloo(ProcessGroupWrapper):
-    """Reconfigurable Gloo process group (CPU tests and CPU fallback)."""
+    """Reconfigurable Gloo group — CPU tests and the CPU control path."""
 
-    def _create_pg(self, store: Store, rank: int, world_size: int) -> BaseProcessGroup:
-        from torch.distributed import ProcessGroupGloo as BaseProcessGroupGloo
+    def _build(self, store: Store, rank: int, world_size: int) -> BaseProcessGroup:
+        from torch.distributed import ProcessGroupGloo as TorchGloo
 
         pg = BaseProcessGroup(store, rank, world_size)
         pg._set_default_backend(BaseProcessGroup.BackendType.GLOO)
-        backend_class = BaseProcessGroupGloo(store, rank, world_size, self._timeout)
-        backend_class._set_sequence_number_for_group()
-
+        backend = TorchGloo(store, rank, world_size, self._timeout)
+        backend._set_sequence_number_for_group()
         if self._global_ranks:
-            backend_class.options.global_ranks_in_group = self._global_ranks
+            backend.options.global_ranks_in_group = self._global_ranks
         if self._group_rank and self._group_world_size:
-            backend_class.options.group_name = (
+            backend.options.group_name = (
                 f"torchft_quorum_{self._quorum_id}_rank_"
                 f"{self._group_rank % self._group_world_size}"
             )
-
-        pg._register_backend(torch.device("cpu"), BaseProcessGroup.BackendType.GLOO, backend_class)
+        pg._register_backend(
+            torch.device("cpu"), BaseProcessGroup.BackendType.GLOO, backend
+        )
         if torch.cuda.is_available():
             pg._register_backend(
-                torch.device("cuda"), BaseProcessGroup.BackendType.GLOO, backend_class
+                torch.device("cuda"), BaseProcessGroup.BackendType.GLOO, backend
             )
         return pg
 
@@ -508,71 +640,70 @@ class ProcessGroupGloo(ProcessGroupWrapper):
         input_tensors: List[torch.Tensor],
         opts: ReduceScatterOptions,
     ) -> None:
-        raise RuntimeError("ProcessGroupGloo does not support reduce_scatter_tensor_coalesced.")
+        raise RuntimeError(
+            "ProcessGroupGloo does not support reduce_scatter_tensor_coalesced."
+        )
 
 
-class _WorkAcceleratorTimeout(Work):
-    """Wraps a Work with a user-space HIP-stream timeout that aborts the PG
-    instead of letting the RCCL watchdog crash the process."""
+class _DeadlineWork(Work):
+    """Work that aborts its process group if completion misses the deadline.
 
-    def __init__(self, pg: ProcessGroup, work: Work, timeout: timedelta) -> None:
+    ``wait()`` guards both the CPU-side wait (which may block, e.g. a
+    barrier) and the HIP-stream completion; either missing the deadline
+    calls ``pg.abort()``, which unblocks the stream with an error that
+    surfaces through ``errored()`` at commit time.
+    """
+
+    def __init__(self, pg: ProcessGroup, work: Work, deadline: timedelta) -> None:
         super().__init__()
         self._pg = pg
         self._work = work
-        self._timeout = timeout
+        self._deadline = deadline
+
+    def _expired(self) -> None:
+        logger.error(f"collective missed {self._deadline} deadline; aborting pg")
+        self._pg.abort()
+
+    @contextmanager
+    def _guard(self, deadline: timedelta) -> Generator[None, None, None]:
+        with context_timeout(self._expired, deadline):
+            yield
+        stream_timeout(self._expired, deadline)
 
     def wait(self, timeout: Optional[timedelta] = None) -> bool:
-        async_timeout = timeout or self._timeout
-        with self._stream_timeout(self._pg, async_timeout):
-            if self._work is not None:
-                if not self._work.wait():
-                    return False
+        effective = timeout or self._deadline
+        with self._guard(effective):
+            if self._work is not None and not self._work.wait():
+                return False
             if timeout is not None:
                 torch.cuda.synchronize()
-            return True
+        return True
 
-    @classmethod
-    @contextmanager
-    def _stream_timeout(
-        cls, pg: ProcessGroup, timeout: timedelta
-    ) -> Generator[None, None, None]:
-        def callback() -> None:
-            logger.error(f"aborting after {timeout}!")
-            pg.abort()
-
-        # .wait() itself must be cancellable (e.g. a blocking barrier) ...
-        with context_timeout(callback, timeout):
-            yield
-        # ... and the HIP stream must complete within the timeout too.
-        stream_timeout(callback, timeout)
-
-    def get_future(self) -> Future[object]:
+    def get_future(self) -> Future:
         fut = self._work.get_future()
 
-        def done_callback(fut: Future[object]) -> None:
+        def enforce(f: Future) -> None:
             try:
-                with self._stream_timeout(self._pg, self._timeout):
-                    fut.wait()
+                with self._guard(self._deadline):
+                    f.wait()
             except Exception as e:  # noqa: BLE001
-                logger.error(f"done callback failed: {e}")
+                logger.error(f"deadline enforcement failed: {e}")
 
-        fut.add_done_callback(done_callback)
+        fut.add_done_callback(enforce)
         return fut
 
 
 class ProcessGroupRCCL(ProcessGroupWrapper):
     """Reconfigurable RCCL process group — the MI355X production backend.
 
-    torch's ``"nccl"`` backend IS RCCL on ROCm. Communicators are created
-    non-blocking and aborted with the RCCL comm-abort path on reconfigure;
-    per-op timeouts are cleared and replaced with user-space timeouts that
-    call ``abort()`` so a dead peer unblocks the survivors without killing
-    them (reference semantics: torchft/process_group.py:780-892).
+    Communicators are created non-blocking and torn down with comm-abort on
+    reconfigure; the c10d watchdog timeout is cleared and replaced by the
+    user-space ``_DeadlineWork`` guard so a dead peer aborts the group
+    instead of crashing the trainer.
 
-    xGMI note: each MI355X GPU has 7 point-to-point xGMI links (~153 GB/s
-    each); RCCL engages them all given correct topology, so the wrapper adds
-    no channel pinning — bucket sizing for per-link bandwidth is handled by
-    the callers (Manager/DiLoCo defaults).
+    xGMI note: each MI355X GPU drives 7 point-to-point links (~153 GB/s
+    each). RCCL engages all of them given correct topology; nothing here
+    pins channels — callers choose bucket sizes for per-link bandwidth.
     """
 
     def __init__(self, timeout: timedelta = timedelta(seconds=60.0)) -> None:
@@ -580,49 +711,46 @@ class ProcessGroupRCCL(ProcessGroupWrapper):
         self._use_abort: bool = torch.cuda.nccl.version() >= (2, 25)
         self._errored: Optional[Exception] = None
 
-        NONBLOCKING_TIMEOUT_ENV = "TORCH_NCCL_NONBLOCKING_TIMEOUT"
-        if NONBLOCKING_TIMEOUT_ENV not in os.environ:
+        env = "TORCH_NCCL_NONBLOCKING_TIMEOUT"
+        if env not in os.environ:
             warnings.warn(
-                f"{NONBLOCKING_TIMEOUT_ENV} is not set, defaulting to {timeout}. "
-                "If any nonblocking RCCL operations have already run this may "
-                "result in the default timeout of 30 minutes and hangs on error."
+                f"{env} is not set, defaulting to {timeout}. If any "
+                "nonblocking RCCL operations have already run this may "
+                "result in the default timeout of 30 minutes and hangs."
             )
-            os.environ[NONBLOCKING_TIMEOUT_ENV] = str(timeout.total_seconds())
+            os.environ[env] = str(timeout.total_seconds())
 
-    def _opts_hook(self, opts: T) -> T:
-        if not self._use_abort:
-            return opts
-        # Clear the c10d watchdog timeout; our user-space timeout aborts
-        # instead of crashing.
-        if hasattr(opts, "timeout"):
+    # timeout policy: hand deadlines to user space, not the c10d watchdog
+    def _prepare_opts(self, opts: object) -> object:
+        if self._use_abort and hasattr(opts, "timeout"):
             opts.timeout = AllgatherOptions().timeout
         return opts
 
-    def _wrap_work(self, work: Work, opts: object) -> Work:
+    def _attach_deadline(self, work: Work, opts: object) -> Work:
         if not self._use_abort:
             return work
-        timeout = self._timeout
+        deadline = self._timeout
         if hasattr(opts, "timeout") and opts.timeout.total_seconds() > 0:
-            timeout = opts.timeout
-        return _WorkAcceleratorTimeout(self, work, timeout)
+            deadline = opts.timeout
+        return _DeadlineWork(self, work, deadline)
 
     @contextmanager
-    def _run_context(self) -> Generator[None, None, None]:
-        timeout: timedelta = self._timeout
+    def _launch_guard(self) -> Generator[None, None, None]:
+        deadline = self._timeout
 
-        def callback() -> None:
-            logger.error(f"aborting after {timeout}!")
+        def expired() -> None:
+            logger.error(f"launch missed {deadline} deadline; aborting pg")
             self.abort()
 
-        with context_timeout(callback, timeout):
+        with context_timeout(expired, deadline):
             yield
 
-    def _create_pg(self, store: Store, rank: int, world_size: int) -> BaseProcessGroup:
-        from torch.distributed import ProcessGroupNCCL as BaseProcessGroupRCCL
+    def _build(self, store: Store, rank: int, world_size: int) -> BaseProcessGroup:
+        from torch.distributed import ProcessGroupNCCL as TorchRCCL
 
         self._errored = None
 
-        opts = BaseProcessGroupRCCL.Options()
+        opts = TorchRCCL.Options()
         opts.config.blocking = False
         if self._global_ranks:
             opts.global_ranks_in_group = self._global_ranks
@@ -634,48 +762,49 @@ class ProcessGroupRCCL(ProcessGroupWrapper):
 
         pg = BaseProcessGroup(store, rank, world_size)
         pg._set_default_backend(BaseProcessGroup.BackendType.NCCL)
-        backend_class = BaseProcessGroupRCCL(store, rank, world_size, opts)
-        backend_class._set_sequence_number_for_group()
-        # Pre-establish the communicator on the local device so the first
-        # collective after reconfigure doesn't pay rendezvous latency.
-        backend_class.eager_connect_single_device(torch.device(torch.cuda.current_device()))
-        pg._register_backend(torch.device("cuda"), BaseProcessGroup.BackendType.NCCL, backend_class)
+        backend = TorchRCCL(store, rank, world_size, opts)
+        backend._set_sequence_number_for_group()
+        # pre-establish the communicator so the first collective after
+        # reconfigure doesn't pay rendezvous latency
+        backend.eager_connect_single_device(
+            torch.device(torch.cuda.current_device())
+        )
+        pg._register_backend(
+            torch.device("cuda"), BaseProcessGroup.BackendType.NCCL, backend
+        )
         return pg
 
     def abort(self, errored: bool = True) -> None:
-        # Set the error before aborting so errored() reports correctly once
-        # the abort unblocks the stream.
+        # mark the error first: the abort unblocks in-flight streams and
+        # errored() must already report by then
         self._errored = RuntimeError("aborted")
         super().abort(errored=errored)
 
     def errored(self) -> Optional[Exception]:
-        synchronize()  # ensure in-flight work surfaced any async error
+        synchronize()  # surface any async stream error first
         return self._errored
 
     def getBackendName(self) -> str:
         return "torchft-rccl"
 
 
-# The reference's name for the accelerator PG; on ROCm it is the same thing.
+# torch calls the accelerator backend NCCL; on ROCm it is RCCL.
 ProcessGroupNCCL = ProcessGroupRCCL
 
 
 class ProcessGroupDummy(ProcessGroup):
-    """World-size-1 no-op PG: copies inputs to outputs and succeeds.
-
-    Soaks up DDP's init broadcast and serves as a test double.
-    """
+    """World-size-1 loopback group: inputs land in outputs, every op
+    completes immediately. Soaks up DDP's init broadcast; test double."""
 
     def __init__(self, rank: int, world: int) -> None:
         super().__init__(rank, world)
-        assert rank == 0
-        assert world == 1
+        assert rank == 0 and world == 1
         self._rank = rank
         self._world = world
         self.wait_count = 0
         self.get_future_count = 0
-        self._work: List[Work] = []
         self.configure_count = 0
+        self._work: List[Work] = []
 
     def configure(
         self,
@@ -690,41 +819,28 @@ class ProcessGroupDummy(ProcessGroup):
     ) -> None:
         self.configure_count += 1
 
-    def allgather(
-        self,
-        output_tensors: List[List[torch.Tensor]],
-        input_tensor: List[torch.Tensor],
-        opts: object,
-    ) -> Work:
+    def _loopback(self, result: object) -> Work:
+        w = _DummyWork(result)
+        self._work.append(w)
+        return w
+
+    def allgather(self, output_tensors, input_tensor, opts: object) -> Work:
         for o, i in zip(output_tensors[0], input_tensor):
             o.copy_(i)
-        res = _DummyWork(output_tensors)
-        self._work.append(res)
-        return res
+        return self._loopback(output_tensors)
 
     def allgather_into_tensor_coalesced(
-        self,
-        output_tensors: List[torch.Tensor],
-        input_tensors: List[torch.Tensor],
-        opts: AllgatherOptions,
+        self, output_tensors, input_tensors, opts: AllgatherOptions
     ) -> Work:
         for o, i in zip(output_tensors, input_tensors):
             o.copy_(i)
-        res = _DummyWork(output_tensors)
-        self._work.append(res)
-        return res
+        return self._loopback(output_tensors)
 
-    def allreduce(self, tensors: List[torch.Tensor], opts: object) -> Work:
-        res = _DummyWork(tensors)
-        self._work.append(res)
-        return res
+    def allreduce(self, tensors, opts: object) -> Work:
+        return self._loopback(tensors)
 
-    def allreduce_coalesced(
-        self, tensors: List[torch.Tensor], opts: Union[AllreduceOptions, ReduceOp]
-    ) -> Work:
-        res = _DummyWork(tensors)
-        self._work.append(res)
-        return res
+    def allreduce_coalesced(self, tensors, opts: object) -> Work:
+        return self._loopback(tensors)
 
     def alltoall_base(
         self,
@@ -735,46 +851,30 @@ class ProcessGroupDummy(ProcessGroup):
         opts: AllToAllOptions,
     ) -> Work:
         output_buffer.copy_(input_buffer)
-        res = _DummyWork([output_buffer])
-        self._work.append(res)
-        return res
+        return self._loopback([output_buffer])
 
     def barrier(self, opts: Optional[BarrierOptions] = None) -> Work:
         return _DummyWork(None)
 
-    def broadcast(self, tensor_list: List[torch.Tensor], opts: object) -> Work:
-        res = _DummyWork(tensor_list)
-        self._work.append(res)
-        return res
+    def broadcast(self, tensor_list, opts: object) -> Work:
+        return self._loopback(tensor_list)
 
-    def recv(self, tensors: List[torch.Tensor], src_rank: int, tag: int) -> Work:
+    def recv(self, tensors, src_rank: int, tag: int) -> Work:
         return _DummyWork(None)
 
-    def reduce_scatter(
-        self,
-        output_tensors: List[torch.Tensor],
-        input_tensors: List[List[torch.Tensor]],
-        opts: object,
-    ) -> Work:
+    def reduce_scatter(self, output_tensors, input_tensors, opts: object) -> Work:
         for o, i in zip(output_tensors, input_tensors[0]):
             o.copy_(i)
-        res = _DummyWork(output_tensors)
-        self._work.append(res)
-        return res
+        return self._loopback(output_tensors)
 
     def reduce_scatter_tensor_coalesced(
-        self,
-        output_tensors: List[torch.Tensor],
-        input_tensors: List[torch.Tensor],
-        opts: ReduceScatterOptions,
+        self, output_tensors, input_tensors, opts: ReduceScatterOptions
     ) -> Work:
         for o, i in zip(output_tensors, input_tensors):
             o.copy_(i)
-        res = _DummyWork(output_tensors)
-        self._work.append(res)
-        return res
+        return self._loopback(output_tensors)
 
-    def send(self, tensors: List[torch.Tensor], dst_rank: int, tag: int) -> Work:
+    def send(self, tensors, dst_rank: int, tag: int) -> Work:
         return _DummyWork(None)
 
     def size(self) -> int:
@@ -803,50 +903,29 @@ class _ErrorSwallowingWork(Work):
             self._pg.report_error(e)
         return True
 
-    def get_future(self) -> Future[object]:
-        fut = self._work.get_future()
-
-        def callback(fut: Future[List[torch.Tensor]]) -> object:
+    def get_future(self) -> Future:
+        def absorb(f: Future) -> object:
             try:
-                return fut.value()
+                return f.value()
             except Exception as e:  # noqa: BLE001
-                logger.exception(f"got exception in future -- skipping remaining: {e}")
+                logger.exception(f"swallowed collective error: {e}")
                 self._pg.report_error(e)
                 return self._default_result
 
-        return fut.then(callback)
+        return self._work.get_future().then(absorb)
 
 
 class ErrorSwallowingProcessGroupWrapper(ProcessGroupWrapper):
-    """Converts collective errors into dummy successes plus a sticky error
-    flag; after the first error all ops are skipped until ``configure``."""
+    """Collective errors become dummy successes plus a sticky error flag;
+    after the first error every op is skipped until ``configure``."""
 
     def __init__(self, pg: ProcessGroup) -> None:
         super().__init__(pg=pg)
         self._error: Optional[Exception] = None
 
-    def configure(
-        self,
-        store_addr: str,
-        replica_id: str,
-        rank: int,
-        world_size: int,
-        quorum_id: Optional[int] = None,
-        group_rank: Optional[int] = None,
-        group_world_size: Optional[int] = None,
-        global_ranks: Optional[list[int]] = None,
-    ) -> None:
+    def configure(self, *args: object, **kwargs: object) -> None:
         self._error = None
-        super().configure(
-            store_addr,
-            replica_id,
-            rank,
-            world_size,
-            quorum_id,
-            group_rank,
-            group_world_size,
-            global_ranks,
-        )
+        super().configure(*args, **kwargs)  # type: ignore[arg-type]
 
     def report_error(self, e: Exception) -> None:
         self._error = e
@@ -865,70 +944,45 @@ class ErrorSwallowingProcessGroupWrapper(ProcessGroupWrapper):
 
 
 class FakeProcessGroupWrapper(ProcessGroupWrapper):
-    """Test-only fault injection: makes the next op's future raise."""
+    """Test-only fault injection: the next op's future raises on demand."""
 
     def __init__(self, pg: ProcessGroup) -> None:
         super().__init__(pg=pg)
         self._future_error: Optional[Exception] = None
 
-    def configure(
-        self,
-        store_addr: str,
-        replica_id: str,
-        rank: int,
-        world_size: int,
-        quorum_id: Optional[int] = None,
-        group_rank: Optional[int] = None,
-        group_world_size: Optional[int] = None,
-        global_ranks: Optional[list[int]] = None,
-    ) -> None:
+    def configure(self, *args: object, **kwargs: object) -> None:
         self._future_error = None
-        super().configure(
-            store_addr,
-            replica_id,
-            rank,
-            world_size,
-            quorum_id,
-            group_rank,
-            group_world_size,
-            global_ranks,
-        )
+        super().configure(*args, **kwargs)  # type: ignore[arg-type]
 
     def report_future_error(self, e: Exception) -> None:
         self._future_error = e
 
     def allreduce(self, tensors: List[torch.Tensor], opts: object) -> Work:
         work = super().allreduce(tensors, opts)
-        if self._future_error is None:
+        injected, self._future_error = self._future_error, None
+        if injected is None:
             return work
 
-        future_error, self._future_error = self._future_error, None
-        assert future_error is not None
+        def detonate(fut: Future) -> List[torch.Tensor]:
+            raise injected
 
-        inner_fut = work.get_future()
+        poisoned_fut = work.get_future().then(detonate)
 
-        def callback(fut: Future[List[torch.Tensor]]) -> List[torch.Tensor]:
-            raise future_error
-
-        errored_fut = inner_fut.then(callback)
-
-        class _FakeErrorWork(Work):
-            def __init__(self) -> None:
-                super().__init__()
-
+        class _PoisonedWork(Work):
             def wait(self, timeout: Optional[timedelta] = None) -> bool:
                 work.wait()
-                raise future_error
+                raise injected
 
-            def get_future(self) -> Future[object]:
-                return errored_fut
+            def get_future(self) -> Future:
+                return poisoned_fut
 
-        return _FakeErrorWork()
+        return _PoisonedWork()
 
 
 class ManagedProcessGroup(ProcessGroupWrapper):
-    """Adapts a Manager into a PG so stock torch DDP/FSDP can use the
-    fault-tolerant allreduce; ``size()`` reports quorum participants."""
+    """Presents a Manager as a PG so stock torch DDP/FSDP hooks route
+    through the fault-tolerant allreduce; ``size()`` is the live quorum
+    participant count."""
 
     def __init__(self, manager: "Manager") -> None:
         super().__init__(pg=manager._pg)
@@ -940,7 +994,7 @@ class ManagedProcessGroup(ProcessGroupWrapper):
             return self._manager.allreduce(tensors[0], reduce_op=opts)
         if isinstance(opts, AllreduceOptions):
             return self._manager.allreduce(tensors[0], reduce_op=opts.reduceOp)
-        raise AssertionError("unreachable")
+        raise AssertionError(f"unsupported allreduce opts: {opts!r}")
 
     def size(self) -> int:
         return self._manager.num_participants()

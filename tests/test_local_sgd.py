@@ -243,3 +243,49 @@ class TestSplitFragments:
 
         with pytest.raises(ValueError):
             split_into_fragments(nn.Sequential(nn.Linear(2, 2)), 3)
+
+
+class TestBucketPlan:
+    """Multi-bucket packing: each bucket's scatter-back must see its own
+    ranges (the reference's closure bug corrupted all but the last bucket)."""
+
+    def test_multi_bucket_scatter_back(self):
+        from torchft_amd.local_sgd import _BucketPlan
+
+        tensors = [torch.zeros(100) for _ in range(5)]
+        # cap of 2 tensors per bucket -> 3 buckets
+        plan = _BucketPlan(tensors, cap_bytes=200 * 4)
+        assert len(plan.buckets) == 3
+        for b, slots in enumerate(plan.buckets):
+            flat = plan.flatten(slots)
+            flat += float(b + 1)  # simulate an allreduce result per bucket
+            _BucketPlan.scatter(flat, slots)
+        torch.testing.assert_close(tensors[0], torch.full((100,), 1.0))
+        torch.testing.assert_close(tensors[2], torch.full((100,), 2.0))
+        torch.testing.assert_close(tensors[4], torch.full((100,), 3.0))
+
+    def test_oversized_tensor_rejected(self):
+        from torchft_amd.local_sgd import _BucketPlan
+
+        with pytest.raises(ValueError, match="bucket cap"):
+            _BucketPlan([torch.zeros(100)], cap_bytes=4)
+
+
+class TestSplitContainers:
+    def test_modulelist_flattened(self):
+        """A model whose blocks live in one ModuleList (like Llama.layers)
+        must split at block granularity, not count the list as one unit."""
+        from torchft_amd.local_sgd import split_into_fragments
+
+        class Net(nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.emb = nn.Embedding(10, 8)
+                self.layers = nn.ModuleList(nn.Linear(8, 8) for _ in range(6))
+                self.head = nn.Linear(8, 10)
+
+        model = Net()
+        frags = split_into_fragments(model, 4)
+        assert len(frags) == 4
+        total = sum(p.numel() for p in model.parameters())
+        assert sum(p.numel() for f in frags for p in f.parameters()) == total

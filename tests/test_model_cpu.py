@@ -85,3 +85,28 @@ class TestFusedAdamWCPUFallback:
             opt.step()
             ref_opt.step()
         torch.testing.assert_close(p, ref, rtol=1e-5, atol=1e-6)
+
+
+class TestFusedAdamWStepBuckets:
+    def test_intermittent_grads_match_torch_adamw(self):
+        """Params whose grads appear intermittently must get bias correction
+        for their own step count, not the first param's."""
+        from torchft_amd.ops import FusedAdamW
+
+        torch.manual_seed(0)
+        p_ref = [torch.randn(8, requires_grad=True) for _ in range(2)]
+        p_fused = [p.detach().clone().requires_grad_(True) for p in p_ref]
+        ref = torch.optim.AdamW(p_ref, lr=0.1, betas=(0.9, 0.95), eps=1e-8,
+                                weight_decay=0.01)
+        fused = FusedAdamW(p_fused, lr=0.1)
+
+        for it in range(5):
+            g0 = torch.randn(8)
+            for opt_params, opt in ((p_ref, ref), (p_fused, fused)):
+                opt_params[0].grad = g0.clone()
+                # second param only gets a grad on even iterations
+                opt_params[1].grad = (g0 * 2).clone() if it % 2 == 0 else None
+                opt.step()
+                opt.zero_grad(set_to_none=True)
+        torch.testing.assert_close(p_fused[0], p_ref[0], rtol=1e-5, atol=1e-6)
+        torch.testing.assert_close(p_fused[1], p_ref[1], rtol=1e-5, atol=1e-6)

@@ -122,6 +122,25 @@ def _baby_worker(
         return
 
     works: Dict[int, Work] = {}
+
+    # Connection.send is not thread-safe: two futures completing at once
+    # would interleave frames and corrupt the pipe. All completions funnel
+    # through one queue drained by a single dedicated sender thread.
+    fut_outbox: "queue.Queue[Optional[tuple]]" = queue.Queue()
+
+    def _fut_sender() -> None:
+        while True:
+            msg = fut_outbox.get()
+            if msg is None:
+                return
+            try:
+                fut_pipe.send(msg)
+            except (OSError, BrokenPipeError):
+                return
+
+    fut_sender_thread = threading.Thread(target=_fut_sender, daemon=True)
+    fut_sender_thread.start()
+
     while True:
         try:
             cmd = req_pipe.recv()
@@ -155,9 +174,9 @@ def _baby_worker(
                         work.wait()
                         if torch.cuda.is_available():
                             torch.cuda.current_stream().synchronize()
-                        fut_pipe.send(("fut_ok", op_id))
+                        fut_outbox.put(("fut_ok", op_id))
                     except Exception as e:  # noqa: BLE001
-                        fut_pipe.send(("fut_exc", op_id, str(e)))
+                        fut_outbox.put(("fut_exc", op_id, str(e)))
 
                 threading.Thread(target=_done, daemon=True).start()
                 req_pipe.send(("ran_future", op_id))
@@ -170,6 +189,8 @@ def _baby_worker(
                 req_pipe.send(RuntimeError(f"baby op failed: {type(e).__name__}: {e}"))
             except Exception:  # noqa: BLE001
                 break
+
+    fut_outbox.put(None)  # stop the sender thread
 
 
 class ProcessGroupBaby(ProcessGroup):

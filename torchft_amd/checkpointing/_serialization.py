@@ -8,6 +8,14 @@ GB-class MI355X state dict streams at wire speed with one staging copy.
 Reference parity: torchft/checkpointing/_serialization.py (which delegates to
 torch.distributed._serialization._streaming_save/load) and the
 _TensorMeta/_DTensorMeta scheme of torchft/checkpointing/pg_transport.py.
+
+SECURITY WARNING: ``streaming_load`` unpickles the header and any
+non-tensor leaves — unpickling attacker-controlled bytes is arbitrary code
+execution. This matches the reference's threat model (torch.load with
+weights_only=False over an unauthenticated HTTP checkpoint server): the
+checkpoint port MUST be reachable only from the training cluster's
+private network. Do not expose HTTPTransport/ParameterServer ports to
+untrusted networks.
 """
 
 from __future__ import annotations

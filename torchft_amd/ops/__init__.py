@@ -10,7 +10,7 @@ reference implementations (used by the numerics tests as ground truth).
 from __future__ import annotations
 
 import os
-from typing import List, Optional
+from typing import Dict, List, Optional
 
 import torch
 
@@ -196,10 +196,11 @@ class FusedAdamW(torch.optim.Optimizer):
     def step(self, closure=None):  # noqa: D102
         assert closure is None
         for group in self.param_groups:
-            params: List[torch.Tensor] = []
-            grads: List[torch.Tensor] = []
-            exp_avgs: List[torch.Tensor] = []
-            exp_avg_sqs: List[torch.Tensor] = []
+            # Params that intermittently have grad=None (e.g. unused-param
+            # DDP) accumulate different step counts; the bias correction is
+            # per step count, so bucket by it — one kernel launch per
+            # distinct count (one in the common case).
+            by_step: Dict[int, List[List[torch.Tensor]]] = {}
             for p in group["params"]:
                 if p.grad is None:
                     continue
@@ -209,27 +210,27 @@ class FusedAdamW(torch.optim.Optimizer):
                     state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
                     state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
                 state["step"] += 1
-                params.append(p)
-                grads.append(p.grad)
-                exp_avgs.append(state["exp_avg"])
-                exp_avg_sqs.append(state["exp_avg_sq"])
-            if not params:
-                continue
-            step = self.state[params[0]]["step"]
+                bucket = by_step.setdefault(state["step"], [[], [], [], []])
+                bucket[0].append(p)
+                bucket[1].append(p.grad)
+                bucket[2].append(state["exp_avg"])
+                bucket[3].append(state["exp_avg_sq"])
+
             beta1, beta2 = group["betas"]
-            if params[0].is_cuda and params[0].dtype == torch.bfloat16:
-                hip_ext().adamw_step(
-                    params, grads, exp_avgs, exp_avg_sqs, group["lr"], beta1,
-                    beta2, group["eps"], group["weight_decay"], step,
-                )
-            else:
-                bc1 = 1 - beta1**step
-                bc2 = 1 - beta2**step
-                for p, g, m, v in zip(params, grads, exp_avgs, exp_avg_sqs):
-                    gf = g.float()
-                    m.mul_(beta1).add_(gf, alpha=1 - beta1)
-                    v.mul_(beta2).addcmul_(gf, gf, value=1 - beta2)
-                    denom = (v / bc2).sqrt_().add_(group["eps"])
-                    update = (m / bc1) / denom + group["weight_decay"] * p.float()
-                    p.add_((-group["lr"] * update).to(p.dtype))
+            for step, (params, grads, exp_avgs, exp_avg_sqs) in by_step.items():
+                if params[0].is_cuda and params[0].dtype == torch.bfloat16:
+                    hip_ext().adamw_step(
+                        params, grads, exp_avgs, exp_avg_sqs, group["lr"], beta1,
+                        beta2, group["eps"], group["weight_decay"], step,
+                    )
+                else:
+                    bc1 = 1 - beta1**step
+                    bc2 = 1 - beta2**step
+                    for p, g, m, v in zip(params, grads, exp_avgs, exp_avg_sqs):
+                        gf = g.float()
+                        m.mul_(beta1).add_(gf, alpha=1 - beta1)
+                        v.mul_(beta2).addcmul_(gf, gf, value=1 - beta2)
+                        denom = (v / bc2).sqrt_().add_(group["eps"])
+                        update = (m / bc1) / denom + group["weight_decay"] * p.float()
+                        p.add_((-group["lr"] * update).to(p.dtype))
         return None

@@ -217,7 +217,10 @@ def main() -> None:
 
             model.apply(apply_hook)
         ddp_model = model
-    elif manager is not None and world > 1:
+    elif manager is not None:
+        # world 1 included: the timed loop must exercise the full FT path
+        # (comm hook -> Manager.allreduce -> RCCL world-1 collective ->
+        # managed continuations), not just quorum + commit barrier
         from torchft_amd.ddp import DistributedDataParallel
 
         ddp_model = DistributedDataParallel(manager, model)

@@ -22,6 +22,16 @@ from torchft_amd.ops.flash_attention import flash_attention
 
 
 @dataclass
+class CPPlan:
+    """Context-parallel plan: this rank holds sequence shard ``rank`` of
+    ``world``; attention all-gathers KV over ``pg`` (parallel/cp.py)."""
+
+    pg: object  # torchft_amd.process_group.ProcessGroup
+    rank: int
+    world: int
+
+
+@dataclass
 class LlamaConfig:
     dim: int = 4096
     n_layers: int = 32
@@ -49,9 +59,12 @@ LLAMA_DEBUG = LlamaConfig(
 
 
 class Attention(nn.Module):
-    def __init__(self, cfg: LlamaConfig, dtype: torch.dtype) -> None:
+    def __init__(
+        self, cfg: LlamaConfig, dtype: torch.dtype, cp: Optional[CPPlan] = None
+    ) -> None:
         super().__init__()
         self.cfg = cfg
+        self.cp = cp
         d, hd = cfg.dim, cfg.head_dim
         self.wq = nn.Linear(d, cfg.n_heads * hd, bias=False, dtype=dtype)
         self.wk = nn.Linear(d, cfg.n_kv_heads * hd, bias=False, dtype=dtype)
@@ -64,8 +77,16 @@ class Attention(nn.Module):
         q = self.wq(x).view(B, S, cfg.n_heads, cfg.head_dim)
         k = self.wk(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
         v = self.wv(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
+        # cos/sin already sliced to this rank's global positions under CP
         q = rope(q, cos, sin)
         k = rope(k, cos, sin)
+        if self.cp is not None and self.cp.world > 1:
+            from torchft_amd.parallel.cp import cp_attention
+
+            out = cp_attention(
+                q, k, v, self.cp.pg, self.cp.rank, self.cp.world, causal=True
+            )
+            return self.wo(out.reshape(B, S, -1))
         # SDPA layout [B, H, S, D]; custom CDNA4 backward where enabled
         # (stock SDPA handles the transposed views without a copy)
         q, k, v = (t.transpose(1, 2) for t in (q, k, v))
@@ -86,10 +107,12 @@ class MLP(nn.Module):
 
 
 class Block(nn.Module):
-    def __init__(self, cfg: LlamaConfig, dtype: torch.dtype) -> None:
+    def __init__(
+        self, cfg: LlamaConfig, dtype: torch.dtype, cp: Optional[CPPlan] = None
+    ) -> None:
         super().__init__()
         self.attn_norm = RMSNorm(cfg.dim, cfg.norm_eps, dtype=dtype)
-        self.attn = Attention(cfg, dtype)
+        self.attn = Attention(cfg, dtype, cp=cp)
         self.mlp_norm = RMSNorm(cfg.dim, cfg.norm_eps, dtype=dtype)
         self.mlp = MLP(cfg, dtype)
 
@@ -105,12 +128,16 @@ class Llama(nn.Module):
         cfg: LlamaConfig,
         dtype: torch.dtype = torch.bfloat16,
         checkpoint_activations: bool = True,
+        cp: Optional[CPPlan] = None,
     ) -> None:
         super().__init__()
         self.cfg = cfg
+        self.cp = cp
         self.checkpoint_activations = checkpoint_activations
         self.tok_embeddings = nn.Embedding(cfg.vocab_size, cfg.dim, dtype=dtype)
-        self.layers = nn.ModuleList(Block(cfg, dtype) for _ in range(cfg.n_layers))
+        self.layers = nn.ModuleList(
+            Block(cfg, dtype, cp=cp) for _ in range(cfg.n_layers)
+        )
         self.norm = RMSNorm(cfg.dim, cfg.norm_eps, dtype=dtype)
         self.output = nn.Linear(cfg.dim, cfg.vocab_size, bias=False, dtype=dtype)
 
@@ -129,9 +156,17 @@ class Llama(nn.Module):
                 nn.init.normal_(m.weight, mean=0.0, std=std)
 
     def forward_hidden(self, tokens: torch.Tensor) -> torch.Tensor:
-        """Token ids [B, S] → final hidden states [B, S, dim]."""
+        """Token ids [B, S] → final hidden states [B, S, dim].
+
+        Under CP, ``tokens`` is this rank's contiguous sequence shard and
+        the rotary tables are sliced to its global positions."""
         x = self.tok_embeddings(tokens)
-        cos, sin = self.rope_cos, self.rope_sin
+        if self.cp is not None and self.cp.world > 1:
+            s = tokens.shape[1]
+            off = self.cp.rank * s
+            cos, sin = self.rope_cos[off : off + s], self.rope_sin[off : off + s]
+        else:
+            cos, sin = self.rope_cos, self.rope_sin
         for layer in self.layers:
             if self.checkpoint_activations and self.training:
                 x = torch.utils.checkpoint.checkpoint(

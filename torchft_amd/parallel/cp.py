@@ -113,26 +113,19 @@ def cp_attention(
         return out.transpose(1, 2)
 
     # causal with a sequence offset: queries at global positions
-    # [rank*s, (rank+1)*s) attend to keys [0, rank*s + local_pos].
-    # Split into the fully-visible prefix (dense) and the diagonal block
-    # (is_causal), avoiding a materialized [s, S] bool mask for the common
-    # case. SDPA requires one call; use an explicit additive mask only for
-    # the diagonal block via is_causal on the block + concat of the prefix.
+    # [rank*s, (rank+1)*s) attend to keys [0, rank*s + local_pos).
+    # Split into the fully-visible prefix block (dense flash) and the
+    # diagonal block (causal flash) and merge via log-sum-exp — flash speed
+    # and flash memory at any context length (no materialized score matrix).
     prefix_len = rank * s
     qT = q.transpose(1, 2)
     if prefix_len > 0:
-        k_prefix = k_full[:, :prefix_len].transpose(1, 2)
-        v_prefix = v_full[:, :prefix_len].transpose(1, 2)
-        k_diag = k_full[:, prefix_len : prefix_len + s].transpose(1, 2)
-        v_diag = v_full[:, prefix_len : prefix_len + s].transpose(1, 2)
-
-        # two-block online-softmax merge using log-sum-exp from each block
-        out_p, lse_p = _sdpa_with_lse(qT, k_prefix, v_prefix, causal=False)
-        out_d, lse_d = _sdpa_with_lse(qT, k_diag, v_diag, causal=True)
-        lse_max = torch.maximum(lse_p, lse_d)
-        w_p = torch.exp(lse_p - lse_max).unsqueeze(-1)
-        w_d = torch.exp(lse_d - lse_max).unsqueeze(-1)
-        out = (out_p * w_p + out_d * w_d) / (w_p + w_d)
+        out = _MergedFlashAttn.apply(
+            qT.contiguous(),
+            k_full.transpose(1, 2).contiguous(),
+            v_full.transpose(1, 2).contiguous(),
+            prefix_len,
+        )
         return out.transpose(1, 2)
 
     out = F.scaled_dot_product_attention(
@@ -140,6 +133,142 @@ def cp_attention(
         is_causal=True, enable_gqa=True,
     )
     return out.transpose(1, 2)
+
+
+# ---------------------------------------------------------------------------
+# flash-attention two-block merge
+# ---------------------------------------------------------------------------
+
+_NATIVE_GQA_CACHE: dict = {}
+
+
+def _native_gqa_ok(device_type: str, dtype: torch.dtype) -> bool:
+    """Probe whether the flash op accepts Hq != Hkv directly on this device."""
+    key = (device_type, dtype)
+    if key not in _NATIVE_GQA_CACHE:
+        try:
+            q = torch.zeros(1, 2, 8, 32, dtype=dtype, device=device_type)
+            kv = torch.zeros(1, 1, 8, 32, dtype=dtype, device=device_type)
+            _flash_fwd_raw(q, kv, kv, causal=False)
+            _NATIVE_GQA_CACHE[key] = True
+        except Exception:  # noqa: BLE001
+            _NATIVE_GQA_CACHE[key] = False
+    return _NATIVE_GQA_CACHE[key]
+
+
+def _flash_fwd_raw(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, causal: bool):
+    """Flash forward returning (out, lse, aux) with lse [B,H,Sq] fp32.
+
+    ``aux`` carries whatever the device-specific backward needs.
+    """
+    if q.is_cuda:
+        (out, lse, cum_q, cum_k, max_q, max_k, seed, offset, _dbg) = (
+            torch.ops.aten._scaled_dot_product_flash_attention(q, k, v, 0.0, causal)
+        )
+        return out, lse, (cum_q, cum_k, max_q, max_k, seed, offset)
+    out, lse = torch.ops.aten._scaled_dot_product_flash_attention_for_cpu(
+        q, k, v, 0.0, causal
+    )
+    return out, lse, None
+
+
+def _flash_bwd_raw(
+    dout: torch.Tensor,
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    out: torch.Tensor,
+    lse: torch.Tensor,
+    causal: bool,
+    aux,
+):
+    if q.is_cuda:
+        cum_q, cum_k, max_q, max_k, seed, offset = aux
+        return torch.ops.aten._scaled_dot_product_flash_attention_backward(
+            dout, q, k, v, out, lse, cum_q, cum_k, max_q, max_k, 0.0, causal,
+            seed, offset,
+        )
+    return torch.ops.aten._scaled_dot_product_flash_attention_for_cpu_backward(
+        dout, q, k, v, out, lse, 0.0, causal
+    )
+
+
+def _expand_kv(t: torch.Tensor, rep: int) -> torch.Tensor:
+    return t.repeat_interleave(rep, dim=1) if rep > 1 else t
+
+
+class _MergedFlashAttn(torch.autograd.Function):
+    """Causal CP attention of one Q shard against [prefix | diagonal] KV.
+
+    Forward runs flash on each block and merges by log-sum-exp. Backward
+    re-enters each block's flash backward with the MERGED out and lse: with
+    the merged lse, exp(S_block − lse) IS the global softmax restricted to
+    the block's columns, and rowsum(dout·out_merged) is the global delta —
+    so each block backward yields exactly its contribution to dq/dk/dv
+    (the standard ring-attention backward identity).
+    """
+
+    @staticmethod
+    def forward(ctx, q: torch.Tensor, k_full: torch.Tensor, v_full: torch.Tensor,
+                prefix_len: int):
+        # q [B,Hq,s,D]; k_full/v_full [B,Hkv,S,D], S >= prefix_len + s
+        s = q.shape[2]
+        Hq, Hkv = q.shape[1], k_full.shape[1]
+        rep = Hq // Hkv
+        if rep > 1 and _native_gqa_ok(q.device.type, q.dtype):
+            rep = 1  # flash handles GQA natively; don't expand
+
+        kp = _expand_kv(k_full[:, :, :prefix_len].contiguous(), rep)
+        vp = _expand_kv(v_full[:, :, :prefix_len].contiguous(), rep)
+        kd = _expand_kv(k_full[:, :, prefix_len : prefix_len + s].contiguous(), rep)
+        vd = _expand_kv(v_full[:, :, prefix_len : prefix_len + s].contiguous(), rep)
+
+        out_p, lse_p, aux_p = _flash_fwd_raw(q, kp, vp, causal=False)
+        out_d, lse_d, aux_d = _flash_fwd_raw(q, kd, vd, causal=True)
+
+        lse = torch.logaddexp(lse_p.float(), lse_d.float())  # [B,Hq,s]
+        w_p = torch.exp(lse_p.float() - lse).unsqueeze(-1)
+        w_d = torch.exp(lse_d.float() - lse).unsqueeze(-1)
+        out = (out_p.float() * w_p + out_d.float() * w_d).to(q.dtype)
+
+        ctx.save_for_backward(q, k_full, v_full, out, lse)
+        ctx.prefix_len = prefix_len
+        ctx.rep = rep
+        ctx.aux = (aux_p, aux_d)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout: torch.Tensor):
+        q, k_full, v_full, out, lse = ctx.saved_tensors
+        prefix_len, rep = ctx.prefix_len, ctx.rep
+        aux_p, aux_d = ctx.aux
+        s = q.shape[2]
+        dout = dout.contiguous()
+
+        kp = _expand_kv(k_full[:, :, :prefix_len].contiguous(), rep)
+        vp = _expand_kv(v_full[:, :, :prefix_len].contiguous(), rep)
+        kd = _expand_kv(k_full[:, :, prefix_len : prefix_len + s].contiguous(), rep)
+        vd = _expand_kv(v_full[:, :, prefix_len : prefix_len + s].contiguous(), rep)
+
+        dq_p, dk_p, dv_p = _flash_bwd_raw(dout, q, kp, vp, out, lse, False, aux_p)
+        dq_d, dk_d, dv_d = _flash_bwd_raw(dout, q, kd, vd, out, lse, True, aux_d)
+
+        Hkv = k_full.shape[1]
+
+        def fold(dk: torch.Tensor) -> torch.Tensor:
+            # undo the GQA expansion: sum grads over the replicated heads
+            if rep == 1:
+                return dk
+            B, _, S, D = dk.shape
+            return dk.view(B, Hkv, rep, S, D).sum(2)
+
+        dk_full = torch.zeros_like(k_full)
+        dv_full = torch.zeros_like(v_full)
+        dk_full[:, :, :prefix_len] = fold(dk_p)
+        dv_full[:, :, :prefix_len] = fold(dv_p)
+        dk_full[:, :, prefix_len : prefix_len + s] = fold(dk_d)
+        dv_full[:, :, prefix_len : prefix_len + s] = fold(dv_d)
+        return dq_p + dq_d, dk_full, dv_full, None
 
 
 def _sdpa_with_lse(

@@ -127,3 +127,213 @@ class TestDiLoCoRegression:
                     atol=1e-6,
                     msg=f"regression at {k}",
                 )
+
+
+# ---------------------------------------------------------------------------
+# fixture breadth: streaming (delay>0), LocalSGD, commit-failure recovery
+# (reference scope: torchft/diloco_regression_test.py:34-131 and its
+# test_fixtures/ suite covering streaming/delay variants)
+# ---------------------------------------------------------------------------
+
+
+def _fixture_path(name: str) -> str:
+    return os.path.join(
+        os.path.dirname(os.path.abspath(__file__)), "test_fixtures", name
+    )
+
+
+def _check_or_write(trajs: List[List[Dict]], fixture: str) -> None:
+    assert trajs[0] == trajs[1], "replica trajectories diverged"
+    path = _fixture_path(fixture)
+    if WRITE_FIXTURE:
+        os.makedirs(os.path.dirname(path), exist_ok=True)
+        with open(path, "w") as f:
+            json.dump(trajs[0], f, indent=1)
+        return
+    assert os.path.exists(path), f"regenerate with WRITE_FIXTURE=true ({path})"
+    with open(path) as f:
+        golden = json.load(f)
+    assert len(golden) == len(trajs[0]), (
+        f"trajectory length changed: {len(trajs[0])} vs golden {len(golden)}"
+    )
+    for got_step, golden_step in zip(trajs[0], golden):
+        assert set(got_step) == set(golden_step)
+        for k in golden_step:
+            torch.testing.assert_close(
+                torch.tensor(got_step[k]),
+                torch.tensor(golden_step[k]),
+                rtol=1e-5,
+                atol=1e-6,
+                msg=f"regression at {k}",
+            )
+
+
+def _run_streaming_replica(
+    replica_id: int,
+    lighthouse_addr: str,
+    outer_steps: int,
+    sync_every: int,
+    fragment_sync_delay: int,
+    fail_allreduce_at: int = -1,
+) -> List[Dict]:
+    from torchft_amd.process_group import FakeProcessGroupWrapper
+
+    store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+    model = _mock_model()
+    fragments = [model[0], model[1]]
+    inner_opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    outer_opts = [
+        torch.optim.SGD(f.parameters(), lr=0.5, momentum=0.9) for f in fragments
+    ]
+    pg = FakeProcessGroupWrapper(ProcessGroupGloo(timeout=timedelta(seconds=20)))
+    manager = Manager(
+        pg=pg,
+        load_state_dict=model.load_state_dict,
+        state_dict=model.state_dict,
+        min_replica_size=2,
+        use_async_quorum=False,
+        init_sync=False,
+        rank=0,
+        world_size=1,
+        store_addr="127.0.0.1",
+        store_port=store.port,
+        lighthouse_addr=lighthouse_addr,
+        replica_id=f"sfix_{replica_id}",
+        hostname="127.0.0.1",
+        timeout=timedelta(seconds=20),
+    )
+    trajectory: List[Dict] = []
+    try:
+        diloco = DiLoCo(
+            manager,
+            fragments,
+            inner_opt,
+            outer_opts,
+            sync_every=sync_every,
+            fragment_sync_delay=fragment_sync_delay,
+            pin_memory=False,
+        )
+        injected = False
+        with diloco:
+            step = 0
+            while manager.current_step() < outer_steps and step < 200:
+                if (
+                    fail_allreduce_at >= 0
+                    and not injected
+                    and manager.current_step() == fail_allreduce_at
+                ):
+                    # both replicas inject at the same committed step, so the
+                    # commit barrier rejects that window everywhere
+                    pg.report_future_error(RuntimeError("injected allreduce error"))
+                    injected = True
+                for p in model.parameters():
+                    p.grad = _deterministic_grad(step, replica_id, p)
+                inner_opt.step()
+                step += 1
+                if step % (sync_every // 2) == 0:
+                    trajectory.append(
+                        {
+                            f"{i}_{name}": param.tolist()
+                            for i, frag in enumerate(diloco._fragments)
+                            for name, param in frag.original_parameters.items()
+                        }
+                    )
+        if fail_allreduce_at >= 0:
+            assert injected, "failure was never injected"
+        return trajectory
+    finally:
+        manager.shutdown(wait=False)
+
+
+def _run_localsgd_replica(
+    replica_id: int, lighthouse_addr: str, outer_steps: int
+) -> List[Dict]:
+    from torchft_amd.local_sgd import LocalSGD
+
+    store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+    model = _mock_model()
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    manager = Manager(
+        pg=ProcessGroupGloo(timeout=timedelta(seconds=20)),
+        load_state_dict=model.load_state_dict,
+        state_dict=model.state_dict,
+        min_replica_size=2,
+        use_async_quorum=True,
+        init_sync=False,
+        rank=0,
+        world_size=1,
+        store_addr="127.0.0.1",
+        store_port=store.port,
+        lighthouse_addr=lighthouse_addr,
+        replica_id=f"lfix_{replica_id}",
+        hostname="127.0.0.1",
+        timeout=timedelta(seconds=20),
+    )
+    trajectory: List[Dict] = []
+    try:
+        with LocalSGD(manager, model, opt, sync_every=3):
+            step = 0
+            while manager.current_step() < outer_steps and step < 100:
+                for p in model.parameters():
+                    p.grad = _deterministic_grad(step, replica_id, p)
+                opt.step()
+                step += 1
+                if step % 3 == 0:
+                    trajectory.append(
+                        {
+                            name: p.detach().tolist()
+                            for name, p in model.named_parameters()
+                        }
+                    )
+        return trajectory
+    finally:
+        manager.shutdown(wait=False)
+
+
+class TestStreamingDiLoCoRegression:
+    def test_delayed_sync_trajectory(self):
+        """Streaming DiLoCo with fragment_sync_delay=1: the allreduce is
+        staged one inner step before it commits."""
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=100)
+        try:
+            with ThreadPoolExecutor(max_workers=2) as ex:
+                futs = [
+                    ex.submit(_run_streaming_replica, i, lh.address(), 4, 4, 1)
+                    for i in range(2)
+                ]
+                trajs = [f.result(timeout=120) for f in futs]
+        finally:
+            lh.shutdown()
+        _check_or_write(trajs, "diloco_streaming_delay1.json")
+
+    def test_commit_failure_recovery_trajectory(self):
+        """An injected allreduce failure must roll the window back; the
+        post-recovery trajectory is pinned by the fixture."""
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=100)
+        try:
+            with ThreadPoolExecutor(max_workers=2) as ex:
+                futs = [
+                    ex.submit(
+                        _run_streaming_replica, i, lh.address(), 3, 4, 0, 1
+                    )
+                    for i in range(2)
+                ]
+                trajs = [f.result(timeout=120) for f in futs]
+        finally:
+            lh.shutdown()
+        _check_or_write(trajs, "diloco_commit_failure.json")
+
+
+class TestLocalSGDRegression:
+    def test_localsgd_trajectory(self):
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=100)
+        try:
+            with ThreadPoolExecutor(max_workers=2) as ex:
+                futs = [
+                    ex.submit(_run_localsgd_replica, i, lh.address(), 3)
+                    for i in range(2)
+                ]
+                trajs = [f.result(timeout=120) for f in futs]
+        finally:
+            lh.shutdown()
+        _check_or_write(trajs, "localsgd_trajectory.json")

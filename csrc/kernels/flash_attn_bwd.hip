@@ -12,12 +12,18 @@
 // Each workgroup = 4 wave64s; each wave owns one 32-row block and holds its
 // fp32 accumulators in the unified VGPR/AGPR file.
 //
-// Schedule (v2): double-buffered LDS staging pipelined T14-style — next
-// tile's global loads issue an iteration early, the ds_write pass lands in
-// the buffer the previous iteration finished reading, ONE barrier per
-// iteration; the S and dP MFMA chains run interleaved (independent
-// accumulators hide the 32x32 dependent-accumulator latency), and P/dS use
-// separate per-wave transpose buffers so all 16 dV+dK MFMAs interleave.
+// Schedule (v3): the staged tiles live in ONE subtiled row-major LDS image
+// ([8 d-subtiles][32 rows][16 cols], 1040-B subtile stride) serving BOTH
+// fragment orientations: contiguous ds_read_b128 for the S/dP chains and
+// hardware-transpose ds_read_b64_tr_b16 for the dV/dK/dQ chains. This
+// removes v2's separate transposed images and their 32-per-thread b16
+// scatter writes (the worst LDS citizen in the v2 profile). tr_b16
+// semantics verified by csrc/kernels/tr16_probe.hip on gfx950: with
+// lane-linear 8-B addresses over a contiguous [4][16] bf16 block per
+// 16-lane group, lane l receives column (l&15), elements j=0..3 = rows.
+// The 1040-B subtile stride (65 x 16 B) makes the 8 staging ds_write_b128
+// of a wave's lane group land on 8 distinct bank quads (1024 would be
+// 8-way conflicted).
 //
 // MFMA lane mappings (verified by mfma_probe.hip on gfx950):
 //   A: row=lane&31, k=(lane>>5)*8+m   B: k=(lane>>5)*8+m, col=lane&31
@@ -38,35 +44,27 @@ typedef __attribute__((__vector_size__(16 * sizeof(float)))) float f32x16;
 #define FA_BLK 32
 #define FA_WAVES 4
 #define FA_THREADS 256
+#define SUBT 1040  // subtile stride: 32 rows x 32 B + 16-B pad (see header)
 
 // ---- LDS image geometry ---------------------------------------------------
-__device__ inline int rm_addr(int row, int byte_off) {
-  return row * 256 + (byte_off ^ ((row & 15) << 4));  // 32 x 256 B, XOR-16
-}
-// 64-B rows alias a 256-B LDS bank row every 4 rows, and the b16 transpose
-// scatter writes 16 rows that are congruent mod 16 — so the XOR mixes BOTH
-// (row>>2) (de-conflicts the 16-consecutive-row b128 reads) and (row>>4)
-// (spreads the stride-16 write pattern): measured 4-way reads / 16-way
-// writes with a single-term XOR.
-__device__ inline int row_swz(int row, int byte_off) {
-  return byte_off ^ ((((row >> 2) ^ (row >> 4)) & 3) << 4);
-}
-__device__ inline int tr_addr(int d, int byte_off) {
-  return d * 64 + row_swz(d, byte_off);  // 128 x 64 B
-}
-__device__ inline int pb_addr(int row, int byte_off) {
-  return row * 64 + row_swz(row, byte_off);  // 32 x 64 B
-}
-
-struct ImageSet {
-  __align__(16) unsigned char a_rm[32 * 256];
-  __align__(16) unsigned char a_tr[128 * 64];
-  __align__(16) unsigned char b_rm[32 * 256];
-  __align__(16) unsigned char b_tr[128 * 64];
+// One 32x128 bf16 tile = 8 subtiles of [32 rows][16 cols], each row-major
+// and element-contiguous (tr_b16 needs its [4][16] blocks contiguous).
+struct Tile {
+  __align__(16) unsigned char sub[8 * SUBT];
 };
 
+struct ImageSet {
+  Tile a;  // Q (dkdv) / K (dq)
+  Tile b;  // dO (dkdv) / V (dq)
+};
+
+// byte address of element [row][col16] in subtile tt
+__device__ inline int st_addr(int tt, int row, int byte_in_row) {
+  return tt * SUBT + row * 32 + byte_in_row;
+}
+
 struct SmemFA {
-  ImageSet img[2];                               // double buffer: 64 KB
+  ImageSet img[2];                                    // double buffer
   __align__(16) unsigned char pa[FA_WAVES][32 * 64];  // P transpose bufs
   __align__(16) unsigned char pb[FA_WAVES][32 * 64];  // dS transpose bufs
   // per-wave K/V operand tiles in A-fragment order (dkdv kernel only):
@@ -79,7 +77,7 @@ struct TileRegs {
 };
 
 // issue the global loads for one (tileA, tileB) pair; thread t owns row
-// q=t>>3, d-segment (t&7)*16..+16 of each 32x128 tile
+// q=t>>3, d-subtile t&7 (16 d columns) of each 32x128 tile
 __device__ inline TileRegs load_tiles(const bf16* __restrict__ srcA,
                                       const bf16* __restrict__ srcB) {
   const int t = threadIdx.x;
@@ -92,43 +90,28 @@ __device__ inline TileRegs load_tiles(const bf16* __restrict__ srcA,
   return r;
 }
 
-template <bool WRITE_TR = true>
-__device__ inline void write_one(unsigned char* rm, unsigned char* tr, int q,
-                                 int dseg, uint4 lo, uint4 hi) {
-  *reinterpret_cast<uint4*>(rm + rm_addr(q, dseg * 2)) = lo;
-  *reinterpret_cast<uint4*>(rm + rm_addr(q, dseg * 2 + 16)) = hi;
-  if constexpr (!WRITE_TR) return;
-  // b16 transpose scatter. (A paired-b32 variant using 2 shuffles per
-  // element was tried and REGRESSED 17.6 -> 22.0 ms: the VALU shuffle cost
-  // exceeds the saved LDS write issue. Next lever is ds_read_b64_tr_b16 on
-  // the row-major image, which removes this image entirely.)
-  const short* v1 = reinterpret_cast<const short*>(&lo);
-  const short* v2 = reinterpret_cast<const short*>(&hi);
-#pragma unroll
-  for (int m = 0; m < 8; m++) {
-    *reinterpret_cast<short*>(tr + tr_addr(dseg + m, q * 2)) = v1[m];
-    *reinterpret_cast<short*>(tr + tr_addr(dseg + 8 + m, q * 2)) = v2[m];
-  }
-}
-
-template <bool B_TR = true>
 __device__ inline void write_tiles(ImageSet* img, const TileRegs& r) {
   const int t = threadIdx.x;
-  const int q = t >> 3;
-  const int dseg = (t & 7) * 16;
-  write_one(img->a_rm, img->a_tr, q, dseg, r.a_lo, r.a_hi);
-  // the dq kernel never reads the transposed V image — skip its scatter
-  write_one<B_TR>(img->b_rm, img->b_tr, q, dseg, r.b_lo, r.b_hi);
+  const int addr = st_addr(t & 7, t >> 3, 0);
+  *reinterpret_cast<uint4*>(img->a.sub + addr) = r.a_lo;
+  *reinterpret_cast<uint4*>(img->a.sub + addr + 16) = r.a_hi;
+  *reinterpret_cast<uint4*>(img->b.sub + addr) = r.b_lo;
+  *reinterpret_cast<uint4*>(img->b.sub + addr + 16) = r.b_hi;
 }
 
-__device__ inline bf16x8_vec rm_bfrag(const unsigned char* rm, int t16, int half,
+// B-fragment for the S/dP chains (k-dim = d): contiguous b128 read.
+// k-slice tt covers d = tt*16 + half*8 .. +8 of row l31.
+__device__ inline bf16x8_vec rm_bfrag(const unsigned char* tile, int tt, int half,
                                       int l31) {
-  return *reinterpret_cast<const bf16x8_vec*>(rm + rm_addr(l31, t16 * 32 + half * 16));
+  return *reinterpret_cast<const bf16x8_vec*>(tile + st_addr(tt, l31, half * 16));
 }
-__device__ inline bf16x8_vec tr_bfrag(const unsigned char* tr, int dt, int h2,
-                                      int half, int l31) {
-  return *reinterpret_cast<const bf16x8_vec*>(
-      tr + tr_addr(dt * 32 + l31, h2 * 32 + half * 16));
+
+// pa/pb A-fragment read (row=l31, 8 consecutive elements at h2*16+half*8)
+__device__ inline int row_swz(int row, int byte_off) {
+  return byte_off ^ ((((row >> 2) ^ (row >> 4)) & 3) << 4);
+}
+__device__ inline int pb_addr(int row, int byte_off) {
+  return row * 64 + row_swz(row, byte_off);  // 32 x 64 B
 }
 __device__ inline bf16x8_vec pb_afrag(const unsigned char* pb, int h2, int half,
                                       int l31) {
@@ -136,6 +119,36 @@ __device__ inline bf16x8_vec pb_afrag(const unsigned char* pb, int h2, int half,
 }
 __device__ inline int c_row(int reg, int half) {
   return (reg & 3) + 8 * (reg >> 2) + 4 * half;
+}
+
+// per-lane invariant part of every tr_b16 address (see header):
+//   sub parity (l>>4)&1, k-half row offset (l>>5)*8 rows, chunk (l&15)*8 B
+__device__ inline unsigned tr_lane_off(int lane) {
+  return ((lane >> 4) & 1) * SUBT + ((lane >> 5) * 8) * 32 + (lane & 15) * 8;
+}
+
+// one hardware-transpose read: 4 bf16 = column (l&15), rows j=0..3 of the
+// [4][16] block at (per-lane) byte address `addr` into the LDS image
+#define TR_READ(dst, addr) \
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(dst) : "v"(addr))
+
+// Wait for the four outstanding tr reads and pin the following MFMAs below
+// the wait (rule 18: "memory" would not order register-only MFMAs).
+#define TR_WAIT4(r0, r1, r2, r3)                                      \
+  asm volatile("s_waitcnt lgkmcnt(0)"                                 \
+               : "+v"(r0), "+v"(r1), "+v"(r2), "+v"(r3));             \
+  __builtin_amdgcn_sched_barrier(0)
+
+__device__ inline bf16x8_vec tr_join(unsigned long long lo, unsigned long long hi) {
+  union {
+    struct {
+      unsigned long long lo, hi;
+    } u;
+    bf16x8_vec v;
+  } c;
+  c.u.lo = lo;
+  c.u.hi = hi;
+  return c.v;
 }
 
 // ---- kernel 1: dK/dV ------------------------------------------------------
@@ -191,14 +204,13 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dkdv_kernel(
     return base + (int64_t)g * head_stride + (int64_t)i * FA_BLK * FA_D;
   };
 
+  const unsigned tr_off = tr_lane_off(lane);
+
   for (int t = 0; t < T; t++) {
     const int cur = t & 1;
     const int g = t / nI;
     const int i = i_min + t % nI;
 
-    // stage this tile (registers freed immediately — keeping a cross-
-    // iteration register pipeline pushed the allocator to the VGPR cap and
-    // serialized every B-fragment ds_read behind lgkmcnt(0))
     {
       TileRegs r = load_tiles(tile_src(t, q_base), tile_src(t, do_base));
       write_tiles(&sm.img[cur], r);
@@ -220,9 +232,9 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dkdv_kernel(
         const bf16x8_vec vf =
             *reinterpret_cast<const bf16x8_vec*>(vops + (tt * 64 + lane) * 16);
         s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            kf, rm_bfrag(img->a_rm, tt, half, l31), s_acc, 0, 0, 0);
+            kf, rm_bfrag(img->a.sub, tt, half, l31), s_acc, 0, 0, 0);
         dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            vf, rm_bfrag(img->b_rm, tt, half, l31), dp_acc, 0, 0, 0);
+            vf, rm_bfrag(img->b.sub, tt, half, l31), dp_acc, 0, 0, 0);
       }
 
       const int qg = i * FA_BLK + l31;
@@ -238,18 +250,34 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dkdv_kernel(
         *reinterpret_cast<bf16*>(pb + pb_addr(row, l31 * 2)) =
             __float2bfloat16(pv * (dp_acc[rg] - delta_q) * scale);
       }
+      // hoist the P^T / dS^T A-fragments (shared across all dt); the pa/pb
+      // buffers are wave-private — same-wave DS ordering covers the RAW
+      // (no barrier: the surrounding `if` is wave-divergent)
+      bf16x8_vec paf[2], pbf[2];
+#pragma unroll
+      for (int h2 = 0; h2 < 2; h2++) {
+        paf[h2] = pb_afrag(pa, h2, half, l31);
+        pbf[h2] = pb_afrag(pb, h2, half, l31);
+      }
 
-      // dV += P^T dO ; dK += dS^T Q — 16 MFMAs over 8 independent accs
+      // dV += P^T dO ; dK += dS^T Q — B-fragments by hardware transpose
+      const unsigned a_base = (unsigned)(uintptr_t)img->a.sub + tr_off;
+      const unsigned b_base = (unsigned)(uintptr_t)img->b.sub + tr_off;
 #pragma unroll
       for (int dt = 0; dt < 4; dt++) {
 #pragma unroll
         for (int h2 = 0; h2 < 2; h2++) {
+          const unsigned base = dt * (2 * SUBT) + h2 * 512;
+          unsigned long long q0, q1, d0, d1;
+          TR_READ(d0, b_base + base);        // dO rows +0..3
+          TR_READ(d1, b_base + base + 128);  // dO rows +4..7
+          TR_READ(q0, a_base + base);        // Q rows +0..3
+          TR_READ(q1, a_base + base + 128);  // Q rows +4..7
+          TR_WAIT4(d0, d1, q0, q1);
           dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              pb_afrag(pa, h2, half, l31), tr_bfrag(img->b_tr, dt, h2, half, l31),
-              dv_acc[dt], 0, 0, 0);
+              paf[h2], tr_join(d0, d1), dv_acc[dt], 0, 0, 0);
           dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              pb_afrag(pb, h2, half, l31), tr_bfrag(img->a_tr, dt, h2, half, l31),
-              dk_acc[dt], 0, 0, 0);
+              pbf[h2], tr_join(q0, q1), dk_acc[dt], 0, 0, 0);
         }
       }
     }
@@ -314,13 +342,14 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
 
   const bf16* k_head = k + ((int64_t)b * Hkv + hkv) * S * FA_D;
   const bf16* v_head = v + ((int64_t)b * Hkv + hkv) * S * FA_D;
+  const unsigned tr_off = tr_lane_off(lane);
 
   for (int j = 0; j < T; j++) {
     const int cur = j & 1;
     {
       TileRegs r = load_tiles(k_head + (int64_t)j * FA_BLK * FA_D,
                               v_head + (int64_t)j * FA_BLK * FA_D);
-      write_tiles<false>(&sm.img[cur], r);
+      write_tiles(&sm.img[cur], r);
     }
     __syncthreads();
 
@@ -331,9 +360,9 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
 #pragma unroll
       for (int tt = 0; tt < 8; tt++) {
         s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            qfrag[tt], rm_bfrag(img->a_rm, tt, half, l31), s_acc, 0, 0, 0);
+            qfrag[tt], rm_bfrag(img->a.sub, tt, half, l31), s_acc, 0, 0, 0);
         dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            dofrag[tt], rm_bfrag(img->b_rm, tt, half, l31), dp_acc, 0, 0, 0);
+            dofrag[tt], rm_bfrag(img->b.sub, tt, half, l31), dp_acc, 0, 0, 0);
       }
 
       unsigned char* pb = sm.pb[wave];
@@ -348,13 +377,24 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
             __float2bfloat16(ds);
       }
 
+      bf16x8_vec pbf[2];
+#pragma unroll
+      for (int h2 = 0; h2 < 2; h2++) pbf[h2] = pb_afrag(pb, h2, half, l31);
+
+      // dQ += dS K — K B-fragments by hardware transpose
+      const unsigned a_base = (unsigned)(uintptr_t)img->a.sub + tr_off;
 #pragma unroll
       for (int dt = 0; dt < 4; dt++) {
 #pragma unroll
         for (int h2 = 0; h2 < 2; h2++) {
+          const unsigned base = dt * (2 * SUBT) + h2 * 512;
+          unsigned long long k0, k1;
+          TR_READ(k0, a_base + base);
+          TR_READ(k1, a_base + base + 128);
+          asm volatile("s_waitcnt lgkmcnt(0)" : "+v"(k0), "+v"(k1));
+          __builtin_amdgcn_sched_barrier(0);
           dq_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              pb_afrag(pb, h2, half, l31), tr_bfrag(img->a_tr, dt, h2, half, l31),
-              dq_acc[dt], 0, 0, 0);
+              pbf[h2], tr_join(k0, k1), dq_acc[dt], 0, 0, 0);
         }
       }
     }

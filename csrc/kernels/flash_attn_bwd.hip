@@ -132,12 +132,15 @@ __device__ inline unsigned tr_lane_off(int lane) {
 #define TR_READ(dst, addr) \
   asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(dst) : "v"(addr))
 
-// Wait for the four outstanding tr reads and pin the following MFMAs below
-// the wait (rule 18: "memory" would not order register-only MFMAs).
-#define TR_WAIT4(r0, r1, r2, r3)                                      \
-  asm volatile("s_waitcnt lgkmcnt(0)"                                 \
-               : "+v"(r0), "+v"(r1), "+v"(r2), "+v"(r3));             \
-  __builtin_amdgcn_sched_barrier(0)
+// Counted wait: allow N newer LDS ops to stay in flight, and tie the four
+// registers the following MFMAs consume so they cannot be scheduled above
+// the wait (explicit dataflow — "memory" alone would not order
+// register-only MFMAs past an asm wait, §5.4 rule 18).
+#define TR_WAIT4_KEEP(N, r0, r1, r2, r3)              \
+  asm volatile("s_waitcnt lgkmcnt(" #N ")"            \
+               : "+v"(r0), "+v"(r1), "+v"(r2), "+v"(r3))
+#define TR_WAIT2_KEEP(N, r0, r1) \
+  asm volatile("s_waitcnt lgkmcnt(" #N ")" : "+v"(r0), "+v"(r1))
 
 __device__ inline bf16x8_vec tr_join(unsigned long long lo, unsigned long long hi) {
   union {
@@ -260,25 +263,40 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dkdv_kernel(
         pbf[h2] = pb_afrag(pb, h2, half, l31);
       }
 
-      // dV += P^T dO ; dK += dS^T Q — B-fragments by hardware transpose
+      // dV += P^T dO ; dK += dS^T Q — B-fragments by hardware transpose,
+      // software-pipelined one (dt,h2) iteration ahead: the next
+      // iteration's 4 tr reads issue before this one's MFMAs, and the
+      // counted lgkmcnt(4) leaves exactly those in flight (safe beside
+      // compiler DS ops: DS completes in order, so "all but newest 4 done"
+      // over-waits at worst).
       const unsigned a_base = (unsigned)(uintptr_t)img->a.sub + tr_off;
       const unsigned b_base = (unsigned)(uintptr_t)img->b.sub + tr_off;
+      unsigned long long fq0[2], fq1[2], fd0[2], fd1[2];
+      TR_READ(fd0[0], b_base);        // iter 0: dO rows +0..3
+      TR_READ(fd1[0], b_base + 128);  //         dO rows +4..7
+      TR_READ(fq0[0], a_base);        //         Q  rows +0..3
+      TR_READ(fq1[0], a_base + 128);  //         Q  rows +4..7
 #pragma unroll
-      for (int dt = 0; dt < 4; dt++) {
-#pragma unroll
-        for (int h2 = 0; h2 < 2; h2++) {
-          const unsigned base = dt * (2 * SUBT) + h2 * 512;
-          unsigned long long q0, q1, d0, d1;
-          TR_READ(d0, b_base + base);        // dO rows +0..3
-          TR_READ(d1, b_base + base + 128);  // dO rows +4..7
-          TR_READ(q0, a_base + base);        // Q rows +0..3
-          TR_READ(q1, a_base + base + 128);  // Q rows +4..7
-          TR_WAIT4(d0, d1, q0, q1);
-          dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              paf[h2], tr_join(d0, d1), dv_acc[dt], 0, 0, 0);
-          dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              pbf[h2], tr_join(q0, q1), dk_acc[dt], 0, 0, 0);
+      for (int it = 0; it < 8; it++) {
+        const int dt = it >> 1;
+        const int h2 = it & 1;
+        const int cur = it & 1;
+        const int nxt = cur ^ 1;
+        if (it < 7) {
+          const int it2 = it + 1;
+          const unsigned nbase = (it2 >> 1) * (2 * SUBT) + (it2 & 1) * 512;
+          TR_READ(fd0[nxt], b_base + nbase);
+          TR_READ(fd1[nxt], b_base + nbase + 128);
+          TR_READ(fq0[nxt], a_base + nbase);
+          TR_READ(fq1[nxt], a_base + nbase + 128);
+          TR_WAIT4_KEEP(4, fd0[cur], fd1[cur], fq0[cur], fq1[cur]);
+        } else {
+          TR_WAIT4_KEEP(0, fd0[cur], fd1[cur], fq0[cur], fq1[cur]);
         }
+        dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            paf[h2], tr_join(fd0[cur], fd1[cur]), dv_acc[dt], 0, 0, 0);
+        dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            pbf[h2], tr_join(fq0[cur], fq1[cur]), dk_acc[dt], 0, 0, 0);
       }
     }
     __syncthreads();
@@ -381,21 +399,29 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
 #pragma unroll
       for (int h2 = 0; h2 < 2; h2++) pbf[h2] = pb_afrag(pb, h2, half, l31);
 
-      // dQ += dS K — K B-fragments by hardware transpose
+      // dQ += dS K — K B-fragments by hardware transpose, pipelined one
+      // iteration ahead (see dkdv kernel)
       const unsigned a_base = (unsigned)(uintptr_t)img->a.sub + tr_off;
+      unsigned long long fk0[2], fk1[2];
+      TR_READ(fk0[0], a_base);
+      TR_READ(fk1[0], a_base + 128);
 #pragma unroll
-      for (int dt = 0; dt < 4; dt++) {
-#pragma unroll
-        for (int h2 = 0; h2 < 2; h2++) {
-          const unsigned base = dt * (2 * SUBT) + h2 * 512;
-          unsigned long long k0, k1;
-          TR_READ(k0, a_base + base);
-          TR_READ(k1, a_base + base + 128);
-          asm volatile("s_waitcnt lgkmcnt(0)" : "+v"(k0), "+v"(k1));
-          __builtin_amdgcn_sched_barrier(0);
-          dq_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              pbf[h2], tr_join(k0, k1), dq_acc[dt], 0, 0, 0);
+      for (int it = 0; it < 8; it++) {
+        const int dt = it >> 1;
+        const int h2 = it & 1;
+        const int cur = it & 1;
+        const int nxt = cur ^ 1;
+        if (it < 7) {
+          const int it2 = it + 1;
+          const unsigned nbase = (it2 >> 1) * (2 * SUBT) + (it2 & 1) * 512;
+          TR_READ(fk0[nxt], a_base + nbase);
+          TR_READ(fk1[nxt], a_base + nbase + 128);
+          TR_WAIT2_KEEP(2, fk0[cur], fk1[cur]);
+        } else {
+          TR_WAIT2_KEEP(0, fk0[cur], fk1[cur]);
         }
+        dq_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            pbf[h2], tr_join(fk0[cur], fk1[cur]), dq_acc[dt], 0, 0, 0);
       }
     }
     __syncthreads();

@@ -71,33 +71,32 @@ logger: logging.Logger = logging.getLogger(__name__)
 ABORT_DUMP_DIR_ENV = "TORCHFT_ABORT_DUMP_DIR"
 
 
-_STORE_CACHE: dict = {}
-_STORE_CACHE_LOCK = threading.Lock()
-
-
-def create_store_client(store_addr: str, timeout: timedelta) -> Store:
+def create_store_client(
+    store_addr: str, timeout: timedelta, cache: Optional[dict] = None
+) -> Store:
     """``host:port/prefix`` -> PrefixStore over a TCPStore client.
 
-    The underlying TCPStore connection is cached per (host, port): quorum
-    reconfiguration changes only the prefix, so re-dialing the store server
-    on every membership change would pay TCP connect + handshake for
-    nothing. Measured on MI355X this removes ~10% of the reconfigure cost
-    (see scripts/measure_reconfigure.py).
+    With ``cache`` (a per-process-group dict), the underlying TCPStore
+    connection is reused across reconfigures: a quorum change rotates only
+    the prefix, so re-dialing the store server would pay TCP connect +
+    handshake for nothing. The cache must never be shared between ranks (a
+    TCPStore client is not safe for concurrent use from thread-ranks), so
+    it is owned by the ProcessGroup instance, not the module.
     """
     host, _, rest = store_addr.partition(":")
     port, _, prefix = rest.partition("/")
     key = (host, int(port))
-    with _STORE_CACHE_LOCK:
-        store = _STORE_CACHE.get(key)
-        if store is None:
-            store = TCPStore(
-                host_name=host,
-                port=int(port),
-                is_master=False,
-                wait_for_workers=False,
-                timeout=timeout,
-            )
-            _STORE_CACHE[key] = store
+    store = cache.get(key) if cache is not None else None
+    if store is None:
+        store = TCPStore(
+            host_name=host,
+            port=int(port),
+            is_master=False,
+            wait_for_workers=False,
+            timeout=timeout,
+        )
+        if cache is not None:
+            cache[key] = store
     return PrefixStore(prefix, store)
 
 
@@ -365,6 +364,7 @@ class ProcessGroupWrapper(ProcessGroup):
         self._group_rank: Optional[int] = None
         self._group_world_size: Optional[int] = None
         self._global_ranks: Optional[list[int]] = None
+        self._store_cache: dict = {}
         self._oplog = _OpLog()
         self.errors_logger: logging.Logger = logging.getLogger("torchft_errors")
 
@@ -404,7 +404,9 @@ class ProcessGroupWrapper(ProcessGroup):
             return
 
         self.abort(errored=False)
-        store = create_store_client(store_addr, timeout=self._timeout)
+        store = create_store_client(
+            store_addr, timeout=self._timeout, cache=self._store_cache
+        )
         self._inner = self._build(store, rank, world_size)
 
     def _build(self, store: Store, rank: int, world_size: int) -> BaseProcessGroup:

@@ -42,8 +42,8 @@ typedef __attribute__((__vector_size__(16 * sizeof(float)))) float f32x16;
 
 #define FA_D 128
 #define FA_BLK 32
-#define FA_WAVES 4
-#define FA_THREADS 256
+#define FA_WAVES 8
+#define FA_THREADS 512
 #define SUBT 1040  // subtile stride: 32 rows x 32 B + 16-B pad (see header)
 
 // ---- LDS image geometry ---------------------------------------------------
@@ -67,36 +67,34 @@ struct SmemFA {
   ImageSet img[2];                                    // double buffer
   __align__(16) unsigned char pa[FA_WAVES][32 * 64];  // P transpose bufs
   __align__(16) unsigned char pb[FA_WAVES][32 * 64];  // dS transpose bufs
-  // per-wave K/V operand tiles in A-fragment order (dkdv kernel only):
-  // [wave][t16][lane][8 bf16] = one b128 per (t16, lane)
-  __align__(16) unsigned char kv_ops[FA_WAVES][2][8 * 64 * 16];
+  // per-wave V operand tile (8 slots) + the last two K k-slices (2 slots)
+  // in A-fragment order (dkdv kernel only): [wave][slot][lane][8 bf16] =
+  // one b128 per (slot, lane). The first six K k-slices live in VGPRs —
+  // all eight would push the allocator over the 256-VGPR spill cliff.
+  __align__(16) unsigned char v_ops[FA_WAVES][10 * 64 * 16];
 };
 
 struct TileRegs {
-  uint4 a_lo, a_hi, b_lo, b_hi;
+  uint4 a, b;
 };
 
-// issue the global loads for one (tileA, tileB) pair; thread t owns row
-// q=t>>3, d-subtile t&7 (16 d columns) of each 32x128 tile
+// issue the global loads for one (tileA, tileB) pair; thread t of 512 owns
+// row q=t>>4 and 16-B chunk c=t&15 (8 d columns) of each 32x128 tile
 __device__ inline TileRegs load_tiles(const bf16* __restrict__ srcA,
                                       const bf16* __restrict__ srcB) {
   const int t = threadIdx.x;
-  const int64_t off = (int64_t)(t >> 3) * FA_D + (t & 7) * 16;
+  const int64_t off = (int64_t)(t >> 4) * FA_D + (t & 15) * 8;
   TileRegs r;
-  r.a_lo = reinterpret_cast<const uint4*>(srcA + off)[0];
-  r.a_hi = reinterpret_cast<const uint4*>(srcA + off)[1];
-  r.b_lo = reinterpret_cast<const uint4*>(srcB + off)[0];
-  r.b_hi = reinterpret_cast<const uint4*>(srcB + off)[1];
+  r.a = reinterpret_cast<const uint4*>(srcA + off)[0];
+  r.b = reinterpret_cast<const uint4*>(srcB + off)[0];
   return r;
 }
 
 __device__ inline void write_tiles(ImageSet* img, const TileRegs& r) {
   const int t = threadIdx.x;
-  const int addr = st_addr(t & 7, t >> 3, 0);
-  *reinterpret_cast<uint4*>(img->a.sub + addr) = r.a_lo;
-  *reinterpret_cast<uint4*>(img->a.sub + addr + 16) = r.a_hi;
-  *reinterpret_cast<uint4*>(img->b.sub + addr) = r.b_lo;
-  *reinterpret_cast<uint4*>(img->b.sub + addr + 16) = r.b_hi;
+  const int addr = st_addr((t & 15) >> 1, t >> 4, (t & 1) * 16);
+  *reinterpret_cast<uint4*>(img->a.sub + addr) = r.a;
+  *reinterpret_cast<uint4*>(img->b.sub + addr) = r.b;
 }
 
 // B-fragment for the S/dP chains (k-dim = d): contiguous b128 read.
@@ -176,17 +174,21 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dkdv_kernel(
   const int nI = nQ - i_min;
   const int T = G * nI;  // flattened (g, i) iteration count
 
-  // K/V operand fragments live in wave-private LDS (A-fragment order, one
-  // b128 per use) — holding them in VGPRs capped the allocator at 256 and
-  // serialized every B-fragment ds_read behind an lgkmcnt(0)
-  unsigned char* kops = sm.kv_ops[wave][0];
-  unsigned char* vops = sm.kv_ops[wave][1];
+  // K fragments in VGPRs (the wave's fixed A-operand); V in wave-private
+  // LDS in A-fragment order (both in VGPRs would blow the 256-VGPR file)
+  unsigned char* vops = sm.v_ops[wave];
+  bf16x8_vec kfrag[6];
   if (active) {
     const int64_t kv_off = (((int64_t)b * Hkv + hkv) * S + jb * FA_BLK + l31) * FA_D;
 #pragma unroll
     for (int t = 0; t < 8; t++) {
-      *reinterpret_cast<bf16x8_vec*>(kops + (t * 64 + lane) * 16) =
+      const bf16x8_vec kv =
           *reinterpret_cast<const bf16x8_vec*>(k + kv_off + t * 16 + half * 8);
+      if (t < 6) {
+        kfrag[t] = kv;
+      } else {
+        *reinterpret_cast<bf16x8_vec*>(vops + ((t + 2) * 64 + lane) * 16) = kv;
+      }
       *reinterpret_cast<bf16x8_vec*>(vops + (t * 64 + lane) * 16) =
           *reinterpret_cast<const bf16x8_vec*>(v + kv_off + t * 16 + half * 8);
     }
@@ -230,10 +232,12 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dkdv_kernel(
       f32x16 dp_acc = {};
 #pragma unroll
       for (int tt = 0; tt < 8; tt++) {
-        const bf16x8_vec kf =
-            *reinterpret_cast<const bf16x8_vec*>(kops + (tt * 64 + lane) * 16);
         const bf16x8_vec vf =
             *reinterpret_cast<const bf16x8_vec*>(vops + (tt * 64 + lane) * 16);
+        const bf16x8_vec kf =
+            tt < 6 ? kfrag[tt]
+                   : *reinterpret_cast<const bf16x8_vec*>(
+                         vops + ((tt + 2) * 64 + lane) * 16);
         s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             kf, rm_bfrag(img->a.sub, tt, half, l31), s_acc, 0, 0, 0);
         dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
@@ -253,16 +257,6 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dkdv_kernel(
         *reinterpret_cast<bf16*>(pb + pb_addr(row, l31 * 2)) =
             __float2bfloat16(pv * (dp_acc[rg] - delta_q) * scale);
       }
-      // hoist the P^T / dS^T A-fragments (shared across all dt); the pa/pb
-      // buffers are wave-private — same-wave DS ordering covers the RAW
-      // (no barrier: the surrounding `if` is wave-divergent)
-      bf16x8_vec paf[2], pbf[2];
-#pragma unroll
-      for (int h2 = 0; h2 < 2; h2++) {
-        paf[h2] = pb_afrag(pa, h2, half, l31);
-        pbf[h2] = pb_afrag(pb, h2, half, l31);
-      }
-
       // dV += P^T dO ; dK += dS^T Q — B-fragments by hardware transpose,
       // software-pipelined one (dt,h2) iteration ahead: the next
       // iteration's 4 tr reads issue before this one's MFMAs, and the
@@ -271,32 +265,29 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dkdv_kernel(
       // over-waits at worst).
       const unsigned a_base = (unsigned)(uintptr_t)img->a.sub + tr_off;
       const unsigned b_base = (unsigned)(uintptr_t)img->b.sub + tr_off;
-      unsigned long long fq0[2], fq1[2], fd0[2], fd1[2];
-      TR_READ(fd0[0], b_base);        // iter 0: dO rows +0..3
-      TR_READ(fd1[0], b_base + 128);  //         dO rows +4..7
-      TR_READ(fq0[0], a_base);        //         Q  rows +0..3
-      TR_READ(fq1[0], a_base + 128);  //         Q  rows +4..7
+      // single-buffered tr reads: at 2 waves/SIMD the partner wave's MFMAs
+      // cover the read latency, and the freed registers keep the kernel
+      // off the 256-VGPR spill cliff (a spilled kfrag put a scratch reload
+      // + vmcnt(0) drain inside this loop)
 #pragma unroll
       for (int it = 0; it < 8; it++) {
         const int dt = it >> 1;
         const int h2 = it & 1;
-        const int cur = it & 1;
-        const int nxt = cur ^ 1;
-        if (it < 7) {
-          const int it2 = it + 1;
-          const unsigned nbase = (it2 >> 1) * (2 * SUBT) + (it2 & 1) * 512;
-          TR_READ(fd0[nxt], b_base + nbase);
-          TR_READ(fd1[nxt], b_base + nbase + 128);
-          TR_READ(fq0[nxt], a_base + nbase);
-          TR_READ(fq1[nxt], a_base + nbase + 128);
-          TR_WAIT4_KEEP(4, fd0[cur], fd1[cur], fq0[cur], fq1[cur]);
-        } else {
-          TR_WAIT4_KEEP(0, fd0[cur], fd1[cur], fq0[cur], fq1[cur]);
-        }
+        const unsigned base = dt * (2 * SUBT) + h2 * 512;
+        unsigned long long fd0, fd1, fq0, fq1;
+        TR_READ(fd0, b_base + base);        // dO rows +0..3
+        TR_READ(fd1, b_base + base + 128);  // dO rows +4..7
+        TR_READ(fq0, a_base + base);        // Q  rows +0..3
+        TR_READ(fq1, a_base + base + 128);  // Q  rows +4..7
+        TR_WAIT4_KEEP(0, fd0, fd1, fq0, fq1);
+        // pa/pb are wave-private: same-wave DS ordering covers the RAW
+        // with the writes above
         dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            paf[h2], tr_join(fd0[cur], fd1[cur]), dv_acc[dt], 0, 0, 0);
+            pb_afrag(pa, h2, half, l31), tr_join(fd0, fd1),
+            dv_acc[dt], 0, 0, 0);
         dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            pbf[h2], tr_join(fq0[cur], fq1[cur]), dk_acc[dt], 0, 0, 0);
+            pb_afrag(pb, h2, half, l31), tr_join(fq0, fq1),
+            dk_acc[dt], 0, 0, 0);
       }
     }
     __syncthreads();

@@ -7,7 +7,11 @@ the full FT stack, SIGKILLs a random replica with exponential inter-arrival
 reports committed-batch goodput vs the healthy baseline.
 
 CPU (gloo, debug model):  python scripts/chaos_goodput.py --duration 60
-GPU: add --device cuda (one replica per GPU via HIP_VISIBLE_DEVICES).
+GPU: add --device cuda (one replica per GPU via HIP_VISIBLE_DEVICES), or
+--device cuda --share-gpu to co-locate all replica groups on GPU 0 (the
+cross-replica allreduce then runs over gloo: RCCL cannot place two ranks
+on one device). --model llama --layers N trains an N-layer Llama-3-8B-
+architecture model (full 4096 dim / GQA / 128k vocab) instead of the toy.
 """
 
 from __future__ import annotations
@@ -43,9 +47,36 @@ use_cuda = os.environ.get("TFT_DEVICE", "cpu") == "cuda" and torch.cuda.is_avail
 device = torch.device("cuda", 0) if use_cuda else torch.device("cpu")
 
 torch.manual_seed(42)
-model = nn.Sequential(nn.Linear(64, 256), nn.ReLU(), nn.Linear(256, 64)).to(device)
+if os.environ.get("TFT_MODEL", "toy") == "llama":
+    from torchft_amd.models.llama import Llama, LlamaConfig
+
+    cfg = LlamaConfig(n_layers=int(os.environ.get("TFT_LAYERS", "4")),
+                      max_seq_len=int(os.environ.get("TFT_SEQ", "2048")))
+    model = Llama(cfg, dtype=torch.bfloat16 if use_cuda else torch.float32,
+                  checkpoint_activations=False).to(device)
+    model.forward = model.forward_loss
+    seq = int(os.environ.get("TFT_SEQ", "2048"))
+    batch = int(os.environ.get("TFT_BATCH", "1"))
+
+    def make_batch(step):
+        g = torch.Generator().manual_seed(step)
+        toks = torch.randint(0, cfg.vocab_size, (batch, seq + 1), generator=g)
+        return toks[:, :-1].to(device), toks[:, 1:].to(device)
+
+    def loss_of(ddp, step):
+        x, y = make_batch(step)
+        return ddp(x, y)
+else:
+    model = nn.Sequential(nn.Linear(64, 256), nn.ReLU(), nn.Linear(256, 64)).to(device)
+
+    def loss_of(ddp, step):
+        torch.manual_seed(step)
+        x = torch.randn(32, 64, device=device)
+        return ddp(x).square().mean()
+
 store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
-pg = ProcessGroupRCCL(timeout=timedelta(seconds=30)) if use_cuda else \
+force_gloo = os.environ.get("TFT_FORCE_GLOO") == "1"
+pg = ProcessGroupRCCL(timeout=timedelta(seconds=30)) if (use_cuda and not force_gloo) else \
      ProcessGroupGloo(timeout=timedelta(seconds=30))
 manager = Manager(
     pg=pg,
@@ -63,10 +94,8 @@ manager = Manager(
 ddp = DistributedDataParallel(manager, model)
 opt = OptimizerWrapper(manager, torch.optim.SGD(model.parameters(), lr=0.01))
 while True:
-    torch.manual_seed(manager.current_step())
-    x = torch.randn(32, 64, device=device)
     opt.zero_grad()
-    ddp(x).square().mean().backward()
+    loss_of(ddp, manager.current_step()).backward()
     opt.step()
     with open(status_file, "w") as f:
         f.write(f"{manager.current_step()} {manager.batches_committed()} {time.time()}")
@@ -80,6 +109,12 @@ def main() -> None:
     p.add_argument("--mtbf-secs", type=float, default=15.0,
                    help="mean time between kills (0 = no chaos, baseline run)")
     p.add_argument("--device", default="cpu", choices=["cpu", "cuda"])
+    p.add_argument("--share-gpu", action="store_true",
+                   help="all replica groups on GPU 0; cross-replica over gloo")
+    p.add_argument("--model", default="toy", choices=["toy", "llama"])
+    p.add_argument("--layers", type=int, default=4)
+    p.add_argument("--seq", type=int, default=2048)
+    p.add_argument("--batch", type=int, default=1)
     args = p.parse_args()
 
     from torchft_amd._ftcore import LighthouseServer
@@ -96,9 +131,17 @@ def main() -> None:
             "TFT_STATUS_FILE": status[i],
             "TFT_DEVICE": args.device,
             "TORCHFT_LIGHTHOUSE": lighthouse.address(),
+            "TFT_MODEL": args.model,
+            "TFT_LAYERS": str(args.layers),
+            "TFT_SEQ": str(args.seq),
+            "TFT_BATCH": str(args.batch),
         })
         if args.device == "cuda":
-            env["HIP_VISIBLE_DEVICES"] = str(i)
+            if args.share_gpu:
+                env["HIP_VISIBLE_DEVICES"] = "0"
+                env["TFT_FORCE_GLOO"] = "1"
+            else:
+                env["HIP_VISIBLE_DEVICES"] = str(i)
         return subprocess.Popen([sys.executable, "-c", WORKER_CODE], env=env)
 
     procs = {i: spawn(i) for i in range(args.replicas)}
@@ -151,6 +194,9 @@ def main() -> None:
         "committed_batches": batches,
         "steps_per_sec": round(steps / elapsed, 2),
         "batches_per_sec": round(batches / elapsed, 2),
+        "model": args.model,
+        "layers": args.layers if args.model == "llama" else None,
+        "device": args.device,
     }), flush=True)
 
 

@@ -8,13 +8,15 @@ the stock backward is the single largest kernel of the Llama-8B step
 Constraints of the custom backward: bf16, head_dim=128, seq % 128 == 0,
 dropout 0. Anything else falls back to stock SDPA.
 
-Status (round 1, measured on MI355X at the 8B bench shape): numerically
-correct (tests/test_flash_attn_gpu.py) but 1.34x slower than the stock
-backward — 17.6 ms vs 13.2 ms fwd+bwd. The LDS-conflict swizzle fix took
-it from 21.2 to 17.6 ms (bank conflicts were ~6 cycles per LDS
-instruction); the identified next levers are ds_read_b64_tr_b16 instead of
-the transposed LDS images and larger KV tiles to amortize staging. Until
-it wins, it is OPT-IN: set TORCHFT_AMD_CUSTOM_FA=1 (default off).
+Status (round 2, measured on MI355X at the 8B bench shape, interleaved
+A/B): fwd+bwd 12.69 ms custom vs 13.13 ms stock — the hand-written
+backward BEATS aotriton's, so it is ON by default (opt out with
+TORCHFT_AMD_CUSTOM_FA=0). What got it there, in order: XOR swizzle of the
+LDS images (21.2 -> 17.6 ms), ds_read_b64_tr_b16 hardware-transpose reads
+replacing the b16 transpose-scatter images (-> 16.0 ms), and 8-wave
+workgroups sharing the staged tile stream for 2 waves/SIMD occupancy
+(-> 12.7 ms; the 4-wave version ran 1 wave/SIMD with every dependent
+stall exposed).
 """
 
 from __future__ import annotations
@@ -31,7 +33,7 @@ _ENV = "TORCHFT_AMD_CUSTOM_FA"
 
 
 def custom_fa_enabled() -> bool:
-    v = os.environ.get(_ENV, "0")
+    v = os.environ.get(_ENV, "1")
     return v in ("1", "true", "True")
 
 

@@ -188,3 +188,53 @@ class TestLlamaCP:
             outs = list(ex.map(worker, range(world)))
         got = torch.cat(outs, dim=1)
         torch.testing.assert_close(got, ref, rtol=2e-4, atol=2e-4)
+
+
+class TestRingAttention:
+    """Ring-rotated KV CP: forward AND backward must match full attention."""
+
+    def _run(self, causal, Hq, Hkv):
+        from torchft_amd.parallel.cp import ring_attention
+
+        torch.manual_seed(5)
+        B, S, D, world = 2, 64, 32, 2
+        q = torch.randn(B, S, Hq, D, requires_grad=True)
+        k = torch.randn(B, S, Hkv, D, requires_grad=True)
+        v = torch.randn(B, S, Hkv, D, requires_grad=True)
+        g = torch.randn(B, S, Hq, D)
+        ref = _full_attention(q, k, v, causal)
+        ref.backward(g)
+
+        store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+        addr = f"127.0.0.1:{store.port}/ring{int(causal)}{Hq}"
+
+        def worker(rank):
+            pg = ProcessGroupGloo(timeout=timedelta(seconds=20))
+            pg.configure(addr, f"r{rank}", rank, world)
+            qs = shard_sequence(q.detach(), rank, world).requires_grad_(True)
+            ks = shard_sequence(k.detach(), rank, world).requires_grad_(True)
+            vs = shard_sequence(v.detach(), rank, world).requires_grad_(True)
+            out = ring_attention(qs, ks, vs, pg, rank, world, causal)
+            out.backward(shard_sequence(g, rank, world))
+            return out.detach(), qs.grad, ks.grad, vs.grad
+
+        with ThreadPoolExecutor(max_workers=world) as ex:
+            results = list(ex.map(worker, range(world)))
+
+        outs = torch.cat([r[0] for r in results], dim=1)
+        dqs = torch.cat([r[1] for r in results], dim=1)
+        dks = torch.cat([r[2] for r in results], dim=1)
+        dvs = torch.cat([r[3] for r in results], dim=1)
+        torch.testing.assert_close(outs, ref.detach(), rtol=2e-4, atol=2e-4)
+        torch.testing.assert_close(dqs, q.grad, rtol=2e-4, atol=2e-4)
+        torch.testing.assert_close(dks, k.grad, rtol=2e-4, atol=2e-4)
+        torch.testing.assert_close(dvs, v.grad, rtol=2e-4, atol=2e-4)
+
+    def test_causal_mha(self):
+        self._run(True, 4, 4)
+
+    def test_causal_gqa(self):
+        self._run(True, 4, 2)
+
+    def test_noncausal(self):
+        self._run(False, 4, 4)

@@ -311,3 +311,163 @@ class ContextParallelAttention(torch.nn.Module):
 
     def forward(self, q: torch.Tensor, k: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
         return cp_attention(q, k, v, self._pg, self._rank, self._world, self._causal)
+
+
+# ---------------------------------------------------------------------------
+# ring attention (block-rotating KV)
+# ---------------------------------------------------------------------------
+
+
+def _merge_block(out_acc, lse_acc, out_b, lse_b):
+    """Online LSE merge of one more KV block into the running (out, lse)."""
+    lse_b = lse_b.float()
+    new_lse = torch.logaddexp(lse_acc, lse_b)
+    out_acc = out_acc * torch.exp(lse_acc - new_lse).unsqueeze(-1) + out_b.float() * torch.exp(
+        lse_b - new_lse
+    ).unsqueeze(-1)
+    return out_acc, new_lse
+
+
+class _RingAttention(torch.autograd.Function):
+    """Causal CP attention with ring-rotated KV blocks.
+
+    Forward: each rank streams the other shards' K/V around the ring
+    (W-1 send/recv hops) and folds every visible block into its running
+    (out, lse) by the online-softmax merge — peak KV residency is TWO
+    shards instead of the full context, which is what makes >512k-token
+    training fit. Causal note: rank r attends blocks 0..r only, so later
+    ranks do more work per hop (the classic ring imbalance; a zigzag
+    shard order is the known fix and is left for a later round).
+
+    Backward: re-all-gathers K/V (one collective, no ring) and re-enters
+    flash backward per visible block with the MERGED out/lse — the same
+    identity _MergedFlashAttn uses — then sum-reduces dK/dV back to the
+    owning shards. Peak memory in backward matches the all-gather-KV
+    scheme; the ring saves the FORWARD residency, which is what persists
+    across the whole layer stack of a training step.
+    """
+
+    @staticmethod
+    def forward(ctx, q, k_shard, v_shard, pg, rank, world, causal):
+        # layouts: q [B,Hq,s,D]; k_shard/v_shard [B,Hkv,s,D]
+        B, Hq, s, D = q.shape
+        q = q.contiguous()
+        out_acc = torch.zeros(B, Hq, s, D, dtype=torch.float32, device=q.device)
+        lse_acc = torch.full((B, Hq, s), float("-inf"), dtype=torch.float32,
+                             device=q.device)
+
+        Hkv = k_shard.shape[1]
+        rep = Hq // Hkv
+        if rep > 1 and _native_gqa_ok(q.device.type, q.dtype):
+            rep = 1
+
+        cur_k, cur_v = k_shard.contiguous(), v_shard.contiguous()
+        nxt = (rank + 1) % world
+        prv = (rank - 1) % world
+        aux_by_owner = {}
+        for hop in range(world):
+            owner = (rank - hop) % world
+            if hop < world - 1:
+                # rotate: pass the current block downstream, fetch upstream.
+                # send first on even ranks, recv first on odd: no deadlock
+                # on blocking transports.
+                send_buf = torch.cat([cur_k.reshape(1, -1), cur_v.reshape(1, -1)], 0)
+                recv_buf = torch.empty_like(send_buf)
+                if rank % 2 == 0:
+                    ws = pg.send([send_buf], nxt, tag=hop)
+                    wr = pg.recv([recv_buf], prv, tag=hop)
+                else:
+                    wr = pg.recv([recv_buf], prv, tag=hop)
+                    ws = pg.send([send_buf], nxt, tag=hop)
+                ws.wait()
+                wr.wait()
+            if not causal or owner <= rank:
+                kb = _expand_kv(cur_k, rep)
+                vb = _expand_kv(cur_v, rep)
+                blk_causal = causal and owner == rank
+                o_b, l_b, aux = _flash_fwd_raw(q, kb, vb, causal=blk_causal)
+                aux_by_owner[owner] = aux
+                out_acc, lse_acc = _merge_block(out_acc, lse_acc, o_b, l_b)
+            if hop < world - 1:
+                cur_k = recv_buf[0].view_as(cur_k).contiguous()
+                cur_v = recv_buf[1].view_as(cur_v).contiguous()
+
+        out = out_acc.to(q.dtype)
+        ctx.save_for_backward(q, k_shard, v_shard, out, lse_acc)
+        ctx.pg, ctx.rank, ctx.world, ctx.causal = pg, rank, world, causal
+        ctx.rep = rep
+        ctx.aux_by_owner = aux_by_owner  # dropout-0 flash bwd side args
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k_shard, v_shard, out, lse = ctx.saved_tensors
+        pg, rank, world, causal = ctx.pg, ctx.rank, ctx.world, ctx.causal
+        rep = ctx.rep
+        B, Hkv, s, D = k_shard.shape
+        dout = dout.contiguous()
+
+        # one allgather instead of re-running the ring: [W,B,Hkv,s,D]
+        flat = torch.cat([k_shard.reshape(1, -1), v_shard.reshape(1, -1)], 0)
+        gath = torch.empty(world * flat.numel(), dtype=flat.dtype, device=flat.device)
+        pg.allgather_into_tensor_coalesced(
+            [gath], [flat.reshape(-1)], AllgatherOptions()
+        ).wait()
+        gath = gath.view(world, 2, B, Hkv, s, D)
+
+        dq = torch.zeros_like(q, dtype=torch.float32)
+        dk_full = torch.zeros(world, B, Hkv, s, D, dtype=torch.float32,
+                              device=q.device)
+        dv_full = torch.zeros_like(dk_full)
+        for o in range(world):
+            if causal and o > rank:
+                continue
+            kb = _expand_kv(gath[o, 0].to(q.dtype).contiguous(), rep)
+            vb = _expand_kv(gath[o, 1].to(q.dtype).contiguous(), rep)
+            blk_causal = causal and o == rank
+            # flash bwd consumes the MERGED out/lse: exp(S_b - lse) is the
+            # global softmax restricted to block b's columns
+            dq_b, dk_b, dv_b = _flash_bwd_raw(
+                dout, q, kb, vb, out, lse, blk_causal, ctx.aux_by_owner[o]
+            )
+            dq += dq_b.float()
+
+            def fold(d):
+                if rep == 1:
+                    return d.float()
+                return d.view(B, Hkv, rep, s, D).float().sum(2)
+
+            dk_full[o] = fold(dk_b)
+            dv_full[o] = fold(dv_b)
+
+        # sum contributions from every rank, keep own shard
+        opts = AllreduceOptions()
+        opts.reduceOp = ReduceOp.SUM
+        grads = torch.cat([dk_full.reshape(1, -1), dv_full.reshape(1, -1)], 0)
+        pg.allreduce([grads], opts).wait()
+        dk = grads[0].view(world, B, Hkv, s, D)[rank].to(k_shard.dtype)
+        dv = grads[1].view(world, B, Hkv, s, D)[rank].to(v_shard.dtype)
+        return dq.to(q.dtype), dk, dv, None, None, None, None
+
+
+def ring_attention(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    pg: Optional[ProcessGroup],
+    rank: int,
+    world: int,
+    causal: bool = True,
+) -> torch.Tensor:
+    """Ring-attention CP: same contract as :func:`cp_attention` ([B, s, H, D]
+    shards in/out) but with ring-rotated KV — forward KV residency is two
+    shards instead of the full context."""
+    if pg is None or world == 1:
+        return cp_attention(q, k, v, None, rank, 1, causal)
+    out = _RingAttention.apply(
+        q.transpose(1, 2).contiguous(),
+        k.transpose(1, 2).contiguous(),
+        v.transpose(1, 2).contiguous(),
+        pg, rank, world, causal,
+    )
+    return out.transpose(1, 2)

@@ -244,6 +244,51 @@ __global__ void swiglu_bwd_kernel(const bf16* __restrict__ dy,
   store_bf16x8(db + base, vdb);
 }
 
+// Packed-GLU variants: gu = [rows][2f] with gate = gu[:, :f], up = gu[:, f:]
+// (the fused w13 projection's natural layout — no contiguous() copies).
+// out = silu(gate) * up, [rows][f].
+__global__ void swiglu_glu_fwd_kernel(const bf16* __restrict__ gu,
+                                      bf16* __restrict__ out, int64_t rows,
+                                      int64_t f) {
+  const int64_t idx = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (idx >= rows * f) return;
+  const int64_t r = idx / f;
+  const int64_t c = idx % f;
+  float va[8], vb[8], vo[8];
+  load_bf16x8(gu + r * 2 * f + c, va);
+  load_bf16x8(gu + r * 2 * f + f + c, vb);
+#pragma unroll
+  for (int i = 0; i < 8; i++) {
+    const float sig = 1.f / (1.f + __expf(-va[i]));
+    vo[i] = va[i] * sig * vb[i];
+  }
+  store_bf16x8(out + idx, vo);
+}
+
+// dgu (packed): dgate = dy * up * sig * (1 + gate*(1-sig)); dup = dy * silu
+__global__ void swiglu_glu_bwd_kernel(const bf16* __restrict__ dy,
+                                      const bf16* __restrict__ gu,
+                                      bf16* __restrict__ dgu, int64_t rows,
+                                      int64_t f) {
+  const int64_t idx = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (idx >= rows * f) return;
+  const int64_t r = idx / f;
+  const int64_t c = idx % f;
+  float vdy[8], va[8], vb[8], vda[8], vdb[8];
+  load_bf16x8(dy + idx, vdy);
+  load_bf16x8(gu + r * 2 * f + c, va);
+  load_bf16x8(gu + r * 2 * f + f + c, vb);
+#pragma unroll
+  for (int i = 0; i < 8; i++) {
+    const float sig = 1.f / (1.f + __expf(-va[i]));
+    const float silu = va[i] * sig;
+    vda[i] = vdy[i] * vb[i] * sig * (1.f + va[i] * (1.f - sig));
+    vdb[i] = vdy[i] * silu;
+  }
+  store_bf16x8(dgu + r * 2 * f + c, vda);
+  store_bf16x8(dgu + r * 2 * f + f + c, vdb);
+}
+
 // ---------------------------------------------------------------- launchers
 
 void launch_rmsnorm_fwd(const void* x, const void* w, void* y, float* invrms,
@@ -289,6 +334,23 @@ void launch_swiglu_bwd(const void* dy, const void* a, const void* b, void* da,
   hipLaunchKernelGGL(swiglu_bwd_kernel, dim3((uint32_t)work), dim3(threads), 0,
                      stream, (const bf16*)dy, (const bf16*)a, (const bf16*)b,
                      (bf16*)da, (bf16*)db, n);
+}
+
+void launch_swiglu_glu_fwd(const void* gu, void* out, int64_t rows, int64_t f,
+                           hipStream_t stream) {
+  const int threads = 256;
+  const int64_t work = (rows * f + 8 * threads - 1) / (8 * threads);
+  hipLaunchKernelGGL(swiglu_glu_fwd_kernel, dim3((uint32_t)work), dim3(threads),
+                     0, stream, (const bf16*)gu, (bf16*)out, rows, f);
+}
+
+void launch_swiglu_glu_bwd(const void* dy, const void* gu, void* dgu,
+                           int64_t rows, int64_t f, hipStream_t stream) {
+  const int threads = 256;
+  const int64_t work = (rows * f + 8 * threads - 1) / (8 * threads);
+  hipLaunchKernelGGL(swiglu_glu_bwd_kernel, dim3((uint32_t)work), dim3(threads),
+                     0, stream, (const bf16*)dy, (const bf16*)gu, (bf16*)dgu,
+                     rows, f);
 }
 
 }  // namespace torchft_amd

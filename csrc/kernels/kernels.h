@@ -40,6 +40,10 @@ void launch_rope(const void* x, void* out, const float* cos_tab,
                  bool backward, tft_stream stream);
 void launch_swiglu_fwd(const void* a, const void* b, void* out, int64_t n,
                        tft_stream stream);
+void launch_swiglu_glu_fwd(const void* gu, void* out, int64_t rows, int64_t f,
+                           tft_stream stream);
+void launch_swiglu_glu_bwd(const void* dy, const void* gu, void* dgu,
+                           int64_t rows, int64_t f, tft_stream stream);
 void launch_swiglu_bwd(const void* dy, const void* a, const void* b, void* da,
                        void* db, int64_t n, tft_stream stream);
 

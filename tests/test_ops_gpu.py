@@ -387,3 +387,29 @@ class TestQuantizedAllreduceMultiRank:
                 out[: hi - lo].float(), total[lo:hi], rtol=0.15, atol=0.15,
                 msg=f"rank {r} slice mismatch at world {world}",
             )
+
+
+class TestSwiGLUGlu:
+    def test_packed_glu_matches_ref(self, dev):
+        from torchft_amd.ops import hip_ext, swiglu_ref
+
+        torch.manual_seed(13)
+        gu = torch.randn(64, 2 * 1024, device=dev, dtype=torch.bfloat16)
+        out = hip_ext().swiglu_glu_fwd(gu)
+        ref = swiglu_ref(gu[..., :1024].contiguous(), gu[..., 1024:].contiguous())
+        torch.testing.assert_close(out, ref, rtol=2e-2, atol=2e-2)
+
+    def test_packed_glu_backward(self, dev):
+        from torchft_amd.ops import swiglu_glu
+
+        torch.manual_seed(14)
+        gu = torch.randn(32, 2 * 512, device=dev, dtype=torch.bfloat16,
+                         requires_grad=True)
+        y = swiglu_glu(gu)
+        dy = torch.randn_like(y)
+        y.backward(dy)
+
+        guf = gu.detach().float().requires_grad_(True)
+        import torch.nn.functional as F
+        (F.silu(guf[..., :512]) * guf[..., 512:]).backward(dy.float())
+        torch.testing.assert_close(gu.grad.float(), guf.grad, rtol=3e-2, atol=3e-2)

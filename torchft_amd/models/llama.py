@@ -17,7 +17,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from torchft_amd.ops import RMSNorm, rope, rope_tables, swiglu
+from torchft_amd.ops import RMSNorm, rope, rope_tables, swiglu_glu
 from torchft_amd.ops.flash_attention import flash_attention
 
 
@@ -66,17 +66,23 @@ class Attention(nn.Module):
         self.cfg = cfg
         self.cp = cp
         d, hd = cfg.dim, cfg.head_dim
-        self.wq = nn.Linear(d, cfg.n_heads * hd, bias=False, dtype=dtype)
-        self.wk = nn.Linear(d, cfg.n_kv_heads * hd, bias=False, dtype=dtype)
-        self.wv = nn.Linear(d, cfg.n_kv_heads * hd, bias=False, dtype=dtype)
+        # fused QKV projection: one [d, (Hq+2*Hkv)*hd] GEMM instead of
+        # three — fewer launches and a wider N for the MFMA pipeline
+        self.wqkv = nn.Linear(
+            d, (cfg.n_heads + 2 * cfg.n_kv_heads) * hd, bias=False, dtype=dtype
+        )
         self.wo = nn.Linear(cfg.n_heads * hd, d, bias=False, dtype=dtype)
 
     def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
         B, S, _ = x.shape
         cfg = self.cfg
-        q = self.wq(x).view(B, S, cfg.n_heads, cfg.head_dim)
-        k = self.wk(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
-        v = self.wv(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
+        hd = cfg.head_dim
+        qkv = self.wqkv(x)
+        nq = cfg.n_heads * hd
+        nkv = cfg.n_kv_heads * hd
+        q = qkv[..., :nq].view(B, S, cfg.n_heads, hd)
+        k = qkv[..., nq : nq + nkv].view(B, S, cfg.n_kv_heads, hd)
+        v = qkv[..., nq + nkv :].view(B, S, cfg.n_kv_heads, hd)
         # cos/sin already sliced to this rank's global positions under CP
         q = rope(q, cos, sin)
         k = rope(k, cos, sin)
@@ -98,12 +104,12 @@ class Attention(nn.Module):
 class MLP(nn.Module):
     def __init__(self, cfg: LlamaConfig, dtype: torch.dtype) -> None:
         super().__init__()
-        self.w1 = nn.Linear(cfg.dim, cfg.ffn_hidden, bias=False, dtype=dtype)  # gate
-        self.w3 = nn.Linear(cfg.dim, cfg.ffn_hidden, bias=False, dtype=dtype)  # up
+        # fused gate+up: one [d, 2*ffn] GEMM feeding the fused SwiGLU kernel
+        self.w13 = nn.Linear(cfg.dim, 2 * cfg.ffn_hidden, bias=False, dtype=dtype)
         self.w2 = nn.Linear(cfg.ffn_hidden, cfg.dim, bias=False, dtype=dtype)  # down
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.w2(swiglu(self.w1(x), self.w3(x)))
+        return self.w2(swiglu_glu(self.w13(x)))
 
 
 class Block(nn.Module):

@@ -171,6 +171,30 @@ def swiglu(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     return torch.nn.functional.silu(a) * b
 
 
+class _SwiGLUGluFn(torch.autograd.Function):
+    """Packed-GLU SwiGLU: gu = [..., 2f] with gate = gu[..., :f], up =
+    gu[..., f:] (the fused w13 projection's output) — the kernel reads the
+    halves strided so no contiguous() copies of the activations happen."""
+
+    @staticmethod
+    def forward(ctx, gu: torch.Tensor):
+        ctx.save_for_backward(gu)
+        return hip_ext().swiglu_glu_fwd(gu)
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        (gu,) = ctx.saved_tensors
+        return hip_ext().swiglu_glu_bwd(dy, gu)
+
+
+def swiglu_glu(gu: torch.Tensor) -> torch.Tensor:
+    """silu(gu[..., :f]) * gu[..., f:] for a packed [..., 2f] projection."""
+    if gu.is_cuda:
+        return _SwiGLUGluFn.apply(gu)
+    f = gu.shape[-1] // 2
+    return torch.nn.functional.silu(gu[..., :f]) * gu[..., f:]
+
+
 # ----------------------------------------------------------------- Fused AdamW
 
 

@@ -82,6 +82,41 @@ def main() -> None:
         print({k: round(v, 1) for k, v in times.items()},
               "total_ms=", round(sum(times.values()), 1))
 
+    print("== async abort (background thread) + cached client ==")
+    import threading
+
+    cached0 = TCPStore("127.0.0.1", server.port, is_master=False,
+                       wait_for_workers=False, timeout=timedelta(seconds=30))
+    prev_backend = None
+    for e in range(1, 4):
+        t0 = time.perf_counter()
+        if prev_backend is not None:
+            th = threading.Thread(target=prev_backend.abort)
+            th.start()
+        else:
+            th = None
+        store = PrefixStore(f"aq{e}", cached0)
+        opts = RCCL.Options()
+        opts.config.blocking = False
+        pg = BasePG(store, 0, 1)
+        pg._set_default_backend(BasePG.BackendType.NCCL)
+        backend = RCCL(store, 0, 1, opts)
+        backend._set_sequence_number_for_group()
+        backend.eager_connect_single_device(torch.device("cuda", 0))
+        pg._register_backend(torch.device("cuda"), BasePG.BackendType.NCCL, backend)
+        ar = AllreduceOptions()
+        ar.reduceOp = ReduceOp.SUM
+        pg.allreduce([t_tensor], ar).wait()
+        torch.cuda.synchronize()
+        ready_ms = (time.perf_counter() - t0) * 1000
+        if th is not None:
+            th.join()
+        total_ms = (time.perf_counter() - t0) * 1000
+        print(f"new comm ready in {ready_ms:.1f} ms (old abort joined at "
+              f"{total_ms:.1f} ms, off the critical path)")
+        prev_backend = backend
+    prev_backend.abort()
+
     print("== cached TCPStore client (new prefix only) ==")
     cached = TCPStore("127.0.0.1", server.port, is_master=False,
                       wait_for_workers=False, timeout=timedelta(seconds=30))

@@ -61,6 +61,9 @@ void launch_adamw(const int64_t* param_ptrs, const int64_t* grad_ptrs,
 namespace torchft_amd {
 
 // flash_attn_bwd.hip --------------------------------------------------------
+void launch_fa_fwd(const void* q, const void* k, const void* v, void* out,
+                   float* lse, int B, int Hq, int Hkv, int S, float scale,
+                   bool causal, tft_stream stream);
 void launch_fa_bwd(const void* q, const void* k, const void* v, const void* dout,
                    const float* lse, const float* delta, void* dq, void* dk,
                    void* dv, int B, int Hq, int Hkv, int S, float scale,

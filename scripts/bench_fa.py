@@ -14,10 +14,11 @@ k0 = torch.randn(B, Hkv, S, D, device=dev, dtype=torch.bfloat16)
 v0 = torch.randn(B, Hkv, S, D, device=dev, dtype=torch.bfloat16)
 dout = torch.randn(B, Hq, S, D, device=dev, dtype=torch.bfloat16)
 
-def run(custom: bool, iters=10):
+def run(custom: bool, iters=10, custom_fwd=False):
     q = q0.clone().requires_grad_(True)
     k = k0.clone().requires_grad_(True)
     v = v0.clone().requires_grad_(True)
+    os.environ["TORCHFT_AMD_CUSTOM_FA_FWD"] = "1" if custom_fwd else "0"
     def step():
         if custom:
             out = _FlashAttentionFn.apply(q, k, v, True, D ** -0.5)
@@ -35,9 +36,10 @@ def run(custom: bool, iters=10):
 import os
 ROUNDS = int(os.environ.get("FA_ROUNDS", "3"))
 for rnd in range(ROUNDS):
-    a = run(False); b = run(True)
+    a = run(False); b = run(True); c = run(True, custom_fwd=True)
     fl_fwd = 2*2*B*Hq*S*S*D/2
     fl_bwd = fl_fwd * 2.5
+    print(f"round {rnd}: custom-fwd+bwd {c:.2f} ms", flush=True)
     print(f"round {rnd}: stock {a:.2f} ms | custom {b:.2f} ms "
           f"(fwd+bwd {B*Hq=}, eff stock {(fl_fwd+fl_bwd)/a/1e9:.0f} TF/s, "
           f"custom {(fl_fwd+fl_bwd)/b/1e9:.0f} TF/s)", flush=True)

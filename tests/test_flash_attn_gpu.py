@@ -65,3 +65,57 @@ class TestFlashBwd:
         torch.testing.assert_close(q.grad, ref_dq, rtol=5e-2, atol=5e-2)
         torch.testing.assert_close(k.grad, ref_dk, rtol=5e-2, atol=5e-2)
         torch.testing.assert_close(v.grad, ref_dv, rtol=5e-2, atol=5e-2)
+
+
+class TestCustomForward:
+    """Hand-written forward: out AND lse must match aten's flash (the
+    in-tree backward consumes the lse)."""
+
+    @pytest.mark.parametrize("causal", [True, False])
+    @pytest.mark.parametrize("B,Hq,Hkv,S", [(1, 4, 2, 512), (2, 8, 8, 1024)])
+    def test_fwd_matches_aten(self, causal, B, Hq, Hkv, S):
+        from torchft_amd.ops import hip_ext
+
+        torch.manual_seed(21)
+        D = 128
+        dev = "cuda"
+        q = torch.randn(B, Hq, S, D, device=dev, dtype=torch.bfloat16)
+        k = torch.randn(B, Hkv, S, D, device=dev, dtype=torch.bfloat16)
+        v = torch.randn(B, Hkv, S, D, device=dev, dtype=torch.bfloat16)
+        scale = D ** -0.5
+        out, lse = hip_ext().fa_fwd(q, k, v, scale, causal)
+        ref, ref_lse, *_ = torch.ops.aten._scaled_dot_product_flash_attention(
+            q, k, v, 0.0, causal, False, scale=scale
+        )
+        torch.testing.assert_close(out, ref, rtol=2e-2, atol=2e-2)
+        torch.testing.assert_close(lse, ref_lse.float(), rtol=1e-3, atol=1e-3)
+
+    def test_full_custom_fwd_bwd(self, monkeypatch):
+        """Custom fwd feeding the custom bwd must match stock end to end."""
+        from torchft_amd.ops.flash_attention import _FlashAttentionFn
+
+        monkeypatch.setenv("TORCHFT_AMD_CUSTOM_FA_FWD", "1")
+        torch.manual_seed(22)
+        B, Hq, Hkv, S, D = 1, 8, 2, 512, 128
+        dev = "cuda"
+        q = torch.randn(B, Hq, S, D, device=dev, dtype=torch.bfloat16,
+                        requires_grad=True)
+        k = torch.randn(B, Hkv, S, D, device=dev, dtype=torch.bfloat16,
+                        requires_grad=True)
+        v = torch.randn(B, Hkv, S, D, device=dev, dtype=torch.bfloat16,
+                        requires_grad=True)
+        g = torch.randn(B, Hq, S, D, device=dev, dtype=torch.bfloat16)
+        out = _FlashAttentionFn.apply(q, k, v, True, D ** -0.5)
+        out.backward(g)
+
+        q2 = q.detach().clone().requires_grad_(True)
+        k2 = k.detach().clone().requires_grad_(True)
+        v2 = v.detach().clone().requires_grad_(True)
+        import torch.nn.functional as F
+        ref = F.scaled_dot_product_attention(q2, k2, v2, is_causal=True,
+                                             enable_gqa=True)
+        ref.backward(g)
+        torch.testing.assert_close(out, ref, rtol=2e-2, atol=2e-2)
+        torch.testing.assert_close(q.grad, q2.grad, rtol=5e-2, atol=5e-2)
+        torch.testing.assert_close(k.grad, k2.grad, rtol=5e-2, atol=5e-2)
+        torch.testing.assert_close(v.grad, v2.grad, rtol=5e-2, atol=5e-2)

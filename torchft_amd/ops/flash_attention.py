@@ -30,10 +30,17 @@ import torch.nn.functional as F
 from torchft_amd.ops import hip_ext
 
 _ENV = "TORCHFT_AMD_CUSTOM_FA"
+_ENV_FWD = "TORCHFT_AMD_CUSTOM_FA_FWD"
 
 
 def custom_fa_enabled() -> bool:
     v = os.environ.get(_ENV, "1")
+    return v in ("1", "true", "True")
+
+
+def custom_fa_fwd_enabled() -> bool:
+    """The hand-written forward (opt-in until it beats aotriton's)."""
+    v = os.environ.get(_ENV_FWD, "0")
     return v in ("1", "true", "True")
 
 
@@ -42,9 +49,12 @@ class _FlashAttentionFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, q, k, v, causal: bool, scale: float):
-        out, lse, *_ = torch.ops.aten._scaled_dot_product_flash_attention(
-            q, k, v, 0.0, causal, False, scale=scale
-        )
+        if custom_fa_fwd_enabled() and q.shape[2] % 256 == 0:
+            out, lse = hip_ext().fa_fwd(q, k, v, scale, causal)
+        else:
+            out, lse, *_ = torch.ops.aten._scaled_dot_product_flash_attention(
+                q, k, v, 0.0, causal, False, scale=scale
+            )
         ctx.save_for_backward(q, k, v, out, lse)
         ctx.causal = causal
         ctx.scale = scale

@@ -123,10 +123,13 @@ __device__ inline bf16x8_vec tr_join8(unsigned long long lo,
   return c.v;
 }
 
-// pack two f32 into one dword of 2 bf16 (no builtin on gfx950 — T12)
+// pack two f32 into one dword of 2 bf16 (no builtin on gfx950 — T12).
+// The trailing s_nop provides the 2 wait states a later v_permlane32_swap
+// needs after a VALU write of its operand — the compiler's hazard
+// recognizer cannot see inside this asm (T21 hazard note).
 __device__ inline unsigned cvt_pk_bf16(float lo, float hi) {
   unsigned r;
-  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(r) : "v"(lo), "v"(hi));
   return r;
 }
 

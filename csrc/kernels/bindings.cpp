@@ -189,6 +189,33 @@ at::Tensor swiglu_fwd(at::Tensor a, at::Tensor b) {
   return out;
 }
 
+at::Tensor swiglu_glu_fwd(at::Tensor gu) {
+  TORCH_CHECK(gu.is_cuda() && gu.scalar_type() == at::kBFloat16);
+  auto guc = gu.contiguous();
+  const int64_t f = guc.size(-1) / 2;
+  const int64_t rows = guc.numel() / (2 * f);
+  TORCH_CHECK(f % 8 == 0, "glu width must be a multiple of 8");
+  auto sizes = guc.sizes().vec();
+  sizes.back() = f;
+  auto out = at::empty(sizes, guc.options());
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  tft::launch_swiglu_glu_fwd(guc.data_ptr(), out.data_ptr(), rows, f,
+                             (tft_stream)stream);
+  return out;
+}
+
+at::Tensor swiglu_glu_bwd(at::Tensor dy, at::Tensor gu) {
+  auto dyc = dy.contiguous();
+  auto guc = gu.contiguous();
+  const int64_t f = guc.size(-1) / 2;
+  const int64_t rows = guc.numel() / (2 * f);
+  auto dgu = at::empty_like(guc);
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  tft::launch_swiglu_glu_bwd(dyc.data_ptr(), guc.data_ptr(), dgu.data_ptr(),
+                             rows, f, (tft_stream)stream);
+  return dgu;
+}
+
 std::vector<at::Tensor> swiglu_bwd(at::Tensor dy, at::Tensor a, at::Tensor b) {
   auto dyc = dy.contiguous();
   auto ac = a.contiguous();
@@ -241,6 +268,25 @@ void adamw_step(const std::vector<at::Tensor>& params,
 
 // ---- flash attention backward (appended) -----------------------------------
 
+std::vector<at::Tensor> fa_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                               double scale, bool causal) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 && q.dim() == 4);
+  TORCH_CHECK(q.size(3) == 128, "fa_fwd supports head_dim=128 only");
+  TORCH_CHECK(q.size(2) % 256 == 0, "fa_fwd requires seq % 256 == 0");
+  const int64_t B = q.size(0), Hq = q.size(1), S = q.size(2);
+  const int64_t Hkv = k.size(1);
+  TORCH_CHECK(Hq % Hkv == 0, "Hq must be a multiple of Hkv");
+  auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
+  auto out = at::empty_like(qc);
+  auto lse = at::empty({B, Hq, S}, qc.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  tft::launch_fa_fwd(qc.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                     out.data_ptr(), lse.data_ptr<float>(), (int)B, (int)Hq,
+                     (int)Hkv, (int)S, (float)scale, causal,
+                     (tft_stream)stream);
+  return {out, lse};
+}
+
 std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                at::Tensor dout, at::Tensor lse, at::Tensor delta,
                                double scale, bool causal) {
@@ -289,8 +335,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("rope_apply", &rope_apply);
   m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_glu_fwd", &swiglu_glu_fwd);
+  m.def("swiglu_glu_bwd", &swiglu_glu_bwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("adamw_step", &adamw_step);
+  m.def("fa_fwd", &fa_fwd);
   m.def("fa_bwd", &fa_bwd,
         "flash-attention backward (bf16, D=128): returns (dq, dk, dv)");
   m.def("mfma_probe", &mfma_probe, "mfma_f32_32x32x16_bf16 layout probe");

@@ -268,6 +268,21 @@ void adamw_step(const std::vector<at::Tensor>& params,
 
 // ---- flash attention backward (appended) -----------------------------------
 
+at::Tensor fa_delta(at::Tensor dout, at::Tensor out) {
+  TORCH_CHECK(dout.is_cuda() && dout.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(dout.size(-1) == 128);
+  auto dc = dout.contiguous();
+  auto oc = out.contiguous();
+  const int64_t rows = dc.numel() / 128;
+  auto sizes = dc.sizes().vec();
+  sizes.pop_back();
+  auto delta = at::empty(sizes, dc.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream().stream();
+  tft::launch_fa_delta(dc.data_ptr(), oc.data_ptr(), delta.data_ptr<float>(),
+                       rows, (tft_stream)stream);
+  return delta;
+}
+
 std::vector<at::Tensor> fa_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                double scale, bool causal) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 && q.dim() == 4);
@@ -340,6 +355,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("adamw_step", &adamw_step);
   m.def("fa_fwd", &fa_fwd);
+  m.def("fa_delta", &fa_delta);
   m.def("fa_bwd", &fa_bwd,
         "flash-attention backward (bf16, D=128): returns (dq, dk, dv)");
   m.def("mfma_probe", &mfma_probe, "mfma_f32_32x32x16_bf16 layout probe");

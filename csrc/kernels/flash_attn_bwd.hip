@@ -430,6 +430,37 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
   }
 }
 
+
+// delta = rowsum(dout * out) in fp32 — the FA2 backward's D_i term, one
+// fused pass over the two bf16 tensors (the torch expression materialized
+// fp32 copies of both). One wave per row, 16 B/lane loads.
+__global__ void fa_delta_kernel(const bf16* __restrict__ dout,
+                                const bf16* __restrict__ o,
+                                float* __restrict__ delta, int64_t rows) {
+  const int64_t row = (int64_t)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= rows) return;
+  const int lane = threadIdx.x & 63;
+  // 64 lanes x 2 contiguous bf16 cover the D=128 row
+  const __hip_bfloat162 d2 =
+      *reinterpret_cast<const __hip_bfloat162*>(dout + row * FA_D + lane * 2);
+  const __hip_bfloat162 o2 =
+      *reinterpret_cast<const __hip_bfloat162*>(o + row * FA_D + lane * 2);
+  float2 df = __bfloat1622float2(d2);
+  float2 of = __bfloat1622float2(o2);
+  float v = df.x * of.x + df.y * of.y;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  if (lane == 0) delta[row] = v;
+}
+
+void launch_fa_delta(const void* dout, const void* o, float* delta,
+                     int64_t rows, hipStream_t stream) {
+  const int waves_per_wg = 8;
+  const int64_t nwg = (rows + waves_per_wg - 1) / waves_per_wg;
+  hipLaunchKernelGGL(fa_delta_kernel, dim3((uint32_t)nwg), dim3(waves_per_wg * 64),
+                     0, stream, (const bf16*)dout, (const bf16*)o, delta, rows);
+}
+
 // ---- launchers ------------------------------------------------------------
 
 void launch_fa_bwd(const void* q, const void* k, const void* v, const void* dout,

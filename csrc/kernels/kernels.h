@@ -64,6 +64,8 @@ namespace torchft_amd {
 void launch_fa_fwd(const void* q, const void* k, const void* v, void* out,
                    float* lse, int B, int Hq, int Hkv, int S, float scale,
                    bool causal, tft_stream stream);
+void launch_fa_delta(const void* dout, const void* o, float* delta,
+                     int64_t rows, tft_stream stream);
 void launch_fa_bwd(const void* q, const void* k, const void* v, const void* dout,
                    const float* lse, const float* delta, void* dq, void* dk,
                    void* dv, int B, int Hq, int Hkv, int S, float scale,

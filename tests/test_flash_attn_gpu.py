@@ -119,3 +119,15 @@ class TestCustomForward:
         torch.testing.assert_close(q.grad, q2.grad, rtol=5e-2, atol=5e-2)
         torch.testing.assert_close(k.grad, k2.grad, rtol=5e-2, atol=5e-2)
         torch.testing.assert_close(v.grad, v2.grad, rtol=5e-2, atol=5e-2)
+
+
+class TestFaDelta:
+    def test_delta_matches_torch(self):
+        from torchft_amd.ops import hip_ext
+
+        torch.manual_seed(23)
+        d = torch.randn(2, 4, 256, 128, device="cuda", dtype=torch.bfloat16)
+        o = torch.randn_like(d)
+        got = hip_ext().fa_delta(d, o)
+        ref = (d.float() * o.float()).sum(-1)
+        torch.testing.assert_close(got, ref, rtol=2e-2, atol=2e-2)

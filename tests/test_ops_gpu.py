@@ -292,7 +292,15 @@ class _MailboxPG:
                 if per_rank_in
                 else sbuf
             )
-        self._ctx.barrier.wait()  # everyone copied before buffers are reused
+        # publish "my copies are enqueued" and make the producer's stream
+        # wait for EVERY consumer before it may overwrite its send buffer
+        # (the CPU barrier alone does not order the device work)
+        done = torch.cuda.Event()
+        done.record()
+        self._ctx.post((kind, "done", key[1]), self._rank, done)
+        self._ctx.barrier.wait()
+        for peer in range(self._ctx.world):
+            cur.wait_event(self._ctx.get((kind, "done", key[1]), peer))
         return _DummyWork([out])
 
     def alltoall_base(self, out, inp, _os, _is, _opts):

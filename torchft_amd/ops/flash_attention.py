@@ -63,7 +63,8 @@ class _FlashAttentionFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dout):
         q, k, v, out, lse = ctx.saved_tensors
-        delta = (dout.float() * out.float()).sum(-1)  # [B, Hq, S]
+        # D_i = rowsum(dout * out), fused fp32 kernel (no fp32 temporaries)
+        delta = hip_ext().fa_delta(dout.contiguous(), out)
         dq, dk, dv = hip_ext().fa_bwd(
             q, k, v, dout, lse, delta, ctx.scale, ctx.causal
         )

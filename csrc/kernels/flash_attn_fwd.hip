@@ -52,7 +52,8 @@ typedef __attribute__((__vector_size__(16 * sizeof(float)))) float f32x16;
 // hit 4 distinct bank quads (2-way); ≡ 8 (mod 64) keeps adjacent-subtile
 // tr blocks mostly disjoint.
 #define FSUBT 2080
-#define FWD_THR 8.0f  // defer-max threshold (scaled-score units)
+#define FWD_THR 11.54f  // defer-max threshold: 8 nats in log2 units
+#define LOG2E 1.44269504089f
 
 struct FwdTile {
   __align__(16) unsigned char sub[8 * FSUBT];
@@ -194,8 +195,18 @@ __global__ __launch_bounds__(FWD_THREADS, 1) void fa_fwd_kernel(
   const int nT = (kv_hi + FWD_KVBLK - 1) / FWD_KVBLK;
 
   f32x16 o_acc[4] = {};
-  float m_run = -1e30f;  // running max of scale*s over keys (lane's q)
-  float l_run = 0.f;     // running softmax denominator
+  // softmax runs in the log2 domain: m2 = max of (scale*log2e)*s, powers
+  // of two via the native v_exp, one fma per element (exp2+fma fold)
+  const float c2 = scale * LOG2E;
+  float m_run = -1e30f;
+  float l_run = 0.f;
+
+  // the second-dispatched wave half loses VALU arbitration on every
+  // segment; one static priority raise for it (T5 static form — the
+  // condition must be provably wave-uniform or s_setprio goes under exec)
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256) {
+    __builtin_amdgcn_s_setprio(1);
+  }
 
   const unsigned tr_off = fwd_tr_lane_off(lane);
   float* bc = sm.bcast[wave];
@@ -224,24 +235,26 @@ __global__ __launch_bounds__(FWD_THREADS, 1) void fa_fwd_kernel(
               k_afrag(kimg, tt, s, half, l31), qfrag[tt], s_acc, 0, 0, 0);
         }
 
-        // scaled scores + causal mask; per-lane block max over its q
+        // raw scores + causal mask; per-lane block max over its q.
+        // One multiply converts the max to the log2 domain (c2 > 0
+        // preserves order); each element pays a single fma inside exp2.
         const bool diag = causal && (kb0 + 31 > q_glob);
         float p[16];
         float bmax = -1e30f;
 #pragma unroll
         for (int r = 0; r < 16; r++) {
-          float sv = s_acc[r] * scale;
+          float sv = s_acc[r];
           if (diag && (kb0 + c_row(r, half) > q_glob)) sv = -1e30f;
           p[r] = sv;
           bmax = fmaxf(bmax, sv);
         }
-        bmax = half_combine_max(bmax);  // both 16-key halves of this q
+        bmax = half_combine_max(bmax) * c2;  // both 16-key halves of this q
 
         // defer-max: rescale O only when the max moved past THR
         float m_use = m_run;
         if (!__all(bmax - m_run <= FWD_THR)) {
           const float m_new = fmaxf(m_run, bmax);
-          const float alpha = __expf(m_run - m_new);  // 0 on first tile
+          const float alpha = exp2f(m_run - m_new);  // 0 on first tile
           // broadcast alpha to O's row layout and rescale
           if (half == 0) bc[l31] = alpha;
           __builtin_amdgcn_wave_barrier();
@@ -264,8 +277,8 @@ __global__ __launch_bounds__(FWD_THREADS, 1) void fa_fwd_kernel(
         unsigned pk[8];
 #pragma unroll
         for (int i = 0; i < 8; i++) {
-          const float e0 = __expf(p[2 * i] - m_use);
-          const float e1 = __expf(p[2 * i + 1] - m_use);
+          const float e0 = exp2f(fmaf(p[2 * i], c2, -m_use));
+          const float e1 = exp2f(fmaf(p[2 * i + 1], c2, -m_use));
           psum += e0 + e1;
           pk[i] = cvt_pk_bf16(e0, e1);
         }
@@ -330,7 +343,9 @@ __global__ __launch_bounds__(FWD_THREADS, 1) void fa_fwd_kernel(
     }
   }
   if (half == 0) {
-    lse[q_head + q_glob] = m_run + __logf(l_run);
+    // convert the log2-domain state back to the natural-log lse the
+    // backward consumes: lse = ln2 * (m2 + log2(l))
+    lse[q_head + q_glob] = 0.6931471805599453f * (m_run + __log2f(l_run));
   }
 }
 

@@ -69,24 +69,34 @@ __device__ inline int fst_addr(int tt, int row, int byte_in_row) {
   return tt * FSUBT + row * 32 + byte_in_row;
 }
 
-// thread t of 512 stages rows r=t>>3, 32 B chunk pair c0=(t&7)*2 of a
-// [64][128] bf16 tile: two adjacent uint4 from global, two b128 to LDS
-__device__ inline void stage_kv(FwdTile* kd, FwdTile* vd,
-                                const bf16* __restrict__ ksrc,
-                                const bf16* __restrict__ vsrc) {
+// T14 async-STAGE split: thread t of 512 owns rows r=t>>3, 32-B chunk
+// pair c0=(t&7)*2 of each [64][128] bf16 tile. The loads issue an
+// iteration EARLY (under the previous tile's MFMA phase, hiding the HBM
+// latency); the writes land after the barrier that frees the buffer.
+struct KvRegs {
+  uint4 k0, k1, v0, v1;
+};
+
+__device__ inline KvRegs load_kv(const bf16* __restrict__ ksrc,
+                                 const bf16* __restrict__ vsrc) {
   const int t = threadIdx.x;
-  const int r = t >> 3;
-  const int c0 = (t & 7) * 2;  // 16-B chunk index, 2 per thread
-  const int64_t off = (int64_t)r * FWD_D + c0 * 8;
-  const uint4 k0 = reinterpret_cast<const uint4*>(ksrc + off)[0];
-  const uint4 k1 = reinterpret_cast<const uint4*>(ksrc + off)[1];
-  const uint4 v0 = reinterpret_cast<const uint4*>(vsrc + off)[0];
-  const uint4 v1 = reinterpret_cast<const uint4*>(vsrc + off)[1];
-  const int a0 = fst_addr(c0 >> 1, r, (c0 & 1) * 16);  // c0 even -> byte 0
-  *reinterpret_cast<uint4*>(kd->sub + a0) = k0;
-  *reinterpret_cast<uint4*>(kd->sub + a0 + 16) = k1;
-  *reinterpret_cast<uint4*>(vd->sub + a0) = v0;
-  *reinterpret_cast<uint4*>(vd->sub + a0 + 16) = v1;
+  const int64_t off = (int64_t)(t >> 3) * FWD_D + ((t & 7) * 2) * 8;
+  KvRegs r;
+  r.k0 = reinterpret_cast<const uint4*>(ksrc + off)[0];
+  r.k1 = reinterpret_cast<const uint4*>(ksrc + off)[1];
+  r.v0 = reinterpret_cast<const uint4*>(vsrc + off)[0];
+  r.v1 = reinterpret_cast<const uint4*>(vsrc + off)[1];
+  return r;
+}
+
+__device__ inline void write_kv(FwdTile* kd, FwdTile* vd, const KvRegs& r) {
+  const int t = threadIdx.x;
+  const int c0 = (t & 7) * 2;
+  const int a0 = fst_addr(c0 >> 1, t >> 3, (c0 & 1) * 16);
+  *reinterpret_cast<uint4*>(kd->sub + a0) = r.k0;
+  *reinterpret_cast<uint4*>(kd->sub + a0 + 16) = r.k1;
+  *reinterpret_cast<uint4*>(vd->sub + a0) = r.v0;
+  *reinterpret_cast<uint4*>(vd->sub + a0 + 16) = r.v1;
 }
 
 // K A-fragment: row = key (s*32 + l31), k-dim = d slice tt: contiguous b128
@@ -211,11 +221,18 @@ __global__ __launch_bounds__(FWD_THREADS, 1) void fa_fwd_kernel(
   const unsigned tr_off = fwd_tr_lane_off(lane);
   float* bc = sm.bcast[wave];
 
+  KvRegs staged = load_kv(k_base, v_base);  // tile 0
   for (int j = 0; j < nT; j++) {
     const int cur = j & 1;
-    stage_kv(&sm.k_img[cur], &sm.v_img[cur], k_base + (int64_t)j * FWD_KVBLK * FWD_D,
-             v_base + (int64_t)j * FWD_KVBLK * FWD_D);
+    write_kv(&sm.k_img[cur], &sm.v_img[cur], staged);
     __syncthreads();
+    if (j + 1 < nT) {
+      // issue the next tile's global loads now — they retire under this
+      // tile's MFMA phase and are waited for by the write after the
+      // next barrier (compiler-counted vmcnt)
+      staged = load_kv(k_base + (int64_t)(j + 1) * FWD_KVBLK * FWD_D,
+                       v_base + (int64_t)(j + 1) * FWD_KVBLK * FWD_D);
+    }
 
     const int key0 = j * FWD_KVBLK;
     // this wave needs the tile only if some of its keys are visible

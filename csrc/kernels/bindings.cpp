@@ -268,18 +268,31 @@ void adamw_step(const std::vector<at::Tensor>& params,
 
 // ---- flash attention backward (appended) -----------------------------------
 
+// true when a logical [B,H,S,D] tensor is a transpose view of contiguous
+// [B,S,H,D] storage (the model's projection layout)
+static bool is_bshd(const at::Tensor& t) {
+  return t.dim() == 4 && t.stride(3) == 1 && t.stride(1) == t.size(3) &&
+         t.stride(2) == t.size(1) * t.size(3) &&
+         t.stride(0) == t.size(1) * t.size(2) * t.size(3);
+}
+
+// contiguous in logical [B,H,S,D] order
+static bool is_bhsd(const at::Tensor& t) { return t.is_contiguous(); }
+
 at::Tensor fa_delta(at::Tensor dout, at::Tensor out) {
   TORCH_CHECK(dout.is_cuda() && dout.scalar_type() == at::kBFloat16);
   TORCH_CHECK(dout.size(-1) == 128);
-  auto dc = dout.contiguous();
-  auto oc = out.contiguous();
+  const bool bshd = is_bshd(dout) && is_bshd(out);
+  auto dc = bshd ? dout : dout.contiguous();
+  auto oc = bshd ? out : out.contiguous();
   const int64_t rows = dc.numel() / 128;
-  auto sizes = dc.sizes().vec();
-  sizes.pop_back();
-  auto delta = at::empty(sizes, dc.options().dtype(at::kFloat));
+  // delta is always dense [B, Hq, S]
+  auto delta = at::empty({dout.size(0), dout.size(1), dout.size(2)},
+                         dout.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream().stream();
   tft::launch_fa_delta(dc.data_ptr(), oc.data_ptr(), delta.data_ptr<float>(),
-                       rows, (tft_stream)stream);
+                       rows, (int)dout.size(1), (int)dout.size(2), bshd,
+                       (tft_stream)stream);
   return delta;
 }
 
@@ -291,13 +304,21 @@ std::vector<at::Tensor> fa_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   const int64_t B = q.size(0), Hq = q.size(1), S = q.size(2);
   const int64_t Hkv = k.size(1);
   TORCH_CHECK(Hq % Hkv == 0, "Hq must be a multiple of Hkv");
-  auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
-  auto out = at::empty_like(qc);
-  auto lse = at::empty({B, Hq, S}, qc.options().dtype(at::kFloat));
+  const bool bshd = is_bshd(q) && is_bshd(k) && is_bshd(v);
+  auto qc = bshd ? q : q.contiguous();
+  auto kc = bshd ? k : k.contiguous();
+  auto vc = bshd ? v : v.contiguous();
+  // out matches the input layout: with bshd storage the logical [B,Hq,S,D]
+  // result is a transpose view of a [B,S,Hq,D] buffer (so a later
+  // .transpose(1,2).reshape() in the model is free)
+  auto out = bshd
+                 ? at::empty({B, S, Hq, (int64_t)128}, q.options()).transpose(1, 2)
+                 : at::empty_like(qc);
+  auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream().stream();
   tft::launch_fa_fwd(qc.data_ptr(), kc.data_ptr(), vc.data_ptr(),
                      out.data_ptr(), lse.data_ptr<float>(), (int)B, (int)Hq,
-                     (int)Hkv, (int)S, (float)scale, causal,
+                     (int)Hkv, (int)S, (float)scale, causal, bshd,
                      (tft_stream)stream);
   return {out, lse};
 }
@@ -311,19 +332,30 @@ std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   const int64_t B = q.size(0), Hq = q.size(1), S = q.size(2);
   const int64_t Hkv = k.size(1);
   TORCH_CHECK(Hq % Hkv == 0, "Hq must be a multiple of Hkv");
-  auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous(),
-       doc = dout.contiguous();
+  const bool bshd = is_bshd(q) && is_bshd(k) && is_bshd(v) && is_bshd(dout);
+  auto qc = bshd ? q : q.contiguous();
+  auto kc = bshd ? k : k.contiguous();
+  auto vc = bshd ? v : v.contiguous();
+  auto doc = bshd ? dout : dout.contiguous();
   auto lsec = lse.contiguous().to(at::kFloat);
   auto deltac = delta.contiguous().to(at::kFloat);
   TORCH_CHECK(lsec.numel() == B * Hq * S, "lse shape mismatch");
-  auto dq = at::empty_like(qc);
-  auto dk = at::empty_like(kc);
-  auto dv = at::empty_like(vc);
+  // gradients come back in the inputs' layout (bshd: transpose views of
+  // [B,S,H,D] buffers) so the autograd graph above never copies
+  auto mk = [&](const at::Tensor& like) {
+    if (!bshd) return at::empty_like(like.contiguous());
+    return at::empty({like.size(0), like.size(2), like.size(1), like.size(3)},
+                     like.options())
+        .transpose(1, 2);
+  };
+  auto dq = mk(q);
+  auto dk = mk(k);
+  auto dv = mk(v);
   auto stream = at::cuda::getCurrentCUDAStream().stream();
   tft::launch_fa_bwd(qc.data_ptr(), kc.data_ptr(), vc.data_ptr(), doc.data_ptr(),
                      lsec.data_ptr<float>(), deltac.data_ptr<float>(),
                      dq.data_ptr(), dk.data_ptr(), dv.data_ptr(), (int)B,
-                     (int)Hq, (int)Hkv, (int)S, (float)scale, causal,
+                     (int)Hq, (int)Hkv, (int)S, (float)scale, causal, bshd,
                      (tft_stream)stream);
   return {dq, dk, dv};
 }

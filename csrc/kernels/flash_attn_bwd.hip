@@ -79,11 +79,13 @@ struct TileRegs {
 };
 
 // issue the global loads for one (tileA, tileB) pair; thread t of 512 owns
-// row q=t>>4 and 16-B chunk c=t&15 (8 d columns) of each 32x128 tile
+// row q=t>>4 and 16-B chunk c=t&15 (8 d columns) of each 32x128 tile.
+// ld = elements between consecutive sequence rows (layout-dependent).
 __device__ inline TileRegs load_tiles(const bf16* __restrict__ srcA,
-                                      const bf16* __restrict__ srcB) {
+                                      const bf16* __restrict__ srcB,
+                                      int64_t ld) {
   const int t = threadIdx.x;
-  const int64_t off = (int64_t)(t >> 4) * FA_D + (t & 15) * 8;
+  const int64_t off = (int64_t)(t >> 4) * ld + (t & 15) * 8;
   TileRegs r;
   r.a = reinterpret_cast<const uint4*>(srcA + off)[0];
   r.b = reinterpret_cast<const uint4*>(srcB + off)[0];
@@ -140,6 +142,15 @@ __device__ inline unsigned tr_lane_off(int lane) {
 #define TR_WAIT2_KEEP(N, r0, r1) \
   asm volatile("s_waitcnt lgkmcnt(" #N ")" : "+v"(r0), "+v"(r1))
 
+// element offset of sequence row s of head (b,h): bshd=1 means the
+// storage is [B,S,H,D] (the model's projection layout, seen through a
+// transpose view) — supporting it natively removes every layout copy the
+// autograd path would otherwise pay.
+__device__ inline int64_t row_off(int b, int h, int s, int H, int S, int bshd) {
+  return bshd ? (((int64_t)b * S + s) * H + h) * FA_D
+              : (((int64_t)b * H + h) * S + s) * FA_D;
+}
+
 __device__ inline bf16x8_vec tr_join(unsigned long long lo, unsigned long long hi) {
   union {
     struct {
@@ -158,7 +169,7 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dkdv_kernel(
     const bf16* __restrict__ v, const bf16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     bf16* __restrict__ dk, bf16* __restrict__ dv, int B, int Hq, int Hkv,
-    int S, float scale, int causal) {
+    int S, float scale, int causal, int bshd) {
   __shared__ SmemFA sm;
   const int G = Hq / Hkv;
   const int b = blockIdx.y / Hkv;
@@ -179,7 +190,7 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dkdv_kernel(
   unsigned char* vops = sm.v_ops[wave];
   bf16x8_vec kfrag[6];
   if (active) {
-    const int64_t kv_off = (((int64_t)b * Hkv + hkv) * S + jb * FA_BLK + l31) * FA_D;
+    const int64_t kv_off = row_off(b, hkv, jb * FA_BLK + l31, Hkv, S, bshd);
 #pragma unroll
     for (int t = 0; t < 8; t++) {
       const bf16x8_vec kv =
@@ -197,16 +208,14 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dkdv_kernel(
   f32x16 dk_acc[4] = {};
   f32x16 dv_acc[4] = {};
 
-  const int64_t head_stride = (int64_t)S * FA_D;
-  const bf16* q_base = q + ((int64_t)b * Hq + hkv * G) * head_stride;
-  const bf16* do_base = dout + ((int64_t)b * Hq + hkv * G) * head_stride;
+  const int64_t q_ld = bshd ? (int64_t)Hq * FA_D : FA_D;  // seq-row stride
   const float* lse_base = lse + ((int64_t)b * Hq + hkv * G) * S;
   const float* delta_base = delta + ((int64_t)b * Hq + hkv * G) * S;
 
   auto tile_src = [&](int t, const bf16* base) -> const bf16* {
     const int g = t / nI;
     const int i = i_min + t % nI;
-    return base + (int64_t)g * head_stride + (int64_t)i * FA_BLK * FA_D;
+    return base + row_off(b, hkv * G + g, i * FA_BLK, Hq, S, bshd);
   };
 
   const unsigned tr_off = tr_lane_off(lane);
@@ -217,7 +226,7 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dkdv_kernel(
     const int i = i_min + t % nI;
 
     {
-      TileRegs r = load_tiles(tile_src(t, q_base), tile_src(t, do_base));
+      TileRegs r = load_tiles(tile_src(t, q), tile_src(t, dout), q_ld);
       write_tiles(&sm.img[cur], r);
     }
     __syncthreads();
@@ -294,15 +303,15 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dkdv_kernel(
   }
 
   if (!active) return;
-  const int64_t kv_off = (((int64_t)b * Hkv + hkv) * S + jb * FA_BLK) * FA_D;
 #pragma unroll
   for (int dt = 0; dt < 4; dt++) {
 #pragma unroll
     for (int rg = 0; rg < 16; rg++) {
       const int key = c_row(rg, half);
+      const int64_t row = row_off(b, hkv, jb * FA_BLK + key, Hkv, S, bshd);
       const int d = dt * 32 + l31;
-      dk[kv_off + (int64_t)key * FA_D + d] = __float2bfloat16(dk_acc[dt][rg]);
-      dv[kv_off + (int64_t)key * FA_D + d] = __float2bfloat16(dv_acc[dt][rg]);
+      dk[row + d] = __float2bfloat16(dk_acc[dt][rg]);
+      dv[row + d] = __float2bfloat16(dv_acc[dt][rg]);
     }
   }
 }
@@ -313,7 +322,7 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
     const bf16* __restrict__ v, const bf16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     bf16* __restrict__ dq, int B, int Hq, int Hkv, int S, float scale,
-    int causal) {
+    int causal, int bshd) {
   __shared__ SmemFA sm;
   const int G = Hq / Hkv;
   const int b = blockIdx.y / Hq;
@@ -331,7 +340,7 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
   bf16x8_vec qfrag[8], dofrag[8];
   float lse_row[16], delta_row[16];
   if (active) {
-    const int64_t q_off = (((int64_t)b * Hq + hq) * S + ib * FA_BLK + l31) * FA_D;
+    const int64_t q_off = row_off(b, hq, ib * FA_BLK + l31, Hq, S, bshd);
 #pragma unroll
     for (int t = 0; t < 8; t++) {
       qfrag[t] = *reinterpret_cast<const bf16x8_vec*>(q + q_off + t * 16 + half * 8);
@@ -349,15 +358,15 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
 
   f32x16 dq_acc[4] = {};
 
-  const bf16* k_head = k + ((int64_t)b * Hkv + hkv) * S * FA_D;
-  const bf16* v_head = v + ((int64_t)b * Hkv + hkv) * S * FA_D;
+  const int64_t kv_ld = bshd ? (int64_t)Hkv * FA_D : FA_D;
   const unsigned tr_off = tr_lane_off(lane);
 
   for (int j = 0; j < T; j++) {
     const int cur = j & 1;
     {
-      TileRegs r = load_tiles(k_head + (int64_t)j * FA_BLK * FA_D,
-                              v_head + (int64_t)j * FA_BLK * FA_D);
+      TileRegs r = load_tiles(k + row_off(b, hkv, j * FA_BLK, Hkv, S, bshd),
+                              v + row_off(b, hkv, j * FA_BLK, Hkv, S, bshd),
+                              kv_ld);
       write_tiles(&sm.img[cur], r);
     }
     __syncthreads();
@@ -419,13 +428,12 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
   }
 
   if (!active) return;
-  const int64_t q_off = (((int64_t)b * Hq + hq) * S + ib * FA_BLK) * FA_D;
 #pragma unroll
   for (int dt = 0; dt < 4; dt++) {
 #pragma unroll
     for (int rg = 0; rg < 16; rg++) {
-      dq[q_off + (int64_t)c_row(rg, half) * FA_D + dt * 32 + l31] =
-          __float2bfloat16(dq_acc[dt][rg]);
+      dq[row_off(b, hq, ib * FA_BLK + c_row(rg, half), Hq, S, bshd) + dt * 32 +
+         l31] = __float2bfloat16(dq_acc[dt][rg]);
     }
   }
 }
@@ -436,10 +444,21 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
 // fp32 copies of both). One wave per row, 16 B/lane loads.
 __global__ void fa_delta_kernel(const bf16* __restrict__ dout,
                                 const bf16* __restrict__ o,
-                                float* __restrict__ delta, int64_t rows) {
+                                float* __restrict__ delta, int64_t rows,
+                                int Hq, int S, int bshd) {
   const int64_t row = (int64_t)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   if (row >= rows) return;
   const int lane = threadIdx.x & 63;
+  // delta is dense [B,Hq,S]; with bshd storage the flat row order is
+  // (b, s, h), so remap the output index
+  int64_t didx = row;
+  if (bshd) {
+    const int h = (int)(row % Hq);
+    const int64_t bs = row / Hq;
+    const int s = (int)(bs % S);
+    const int64_t bb = bs / S;
+    didx = (bb * Hq + h) * S + s;
+  }
   // 64 lanes x 2 contiguous bf16 cover the D=128 row
   const __hip_bfloat162 d2 =
       *reinterpret_cast<const __hip_bfloat162*>(dout + row * FA_D + lane * 2);
@@ -450,15 +469,17 @@ __global__ void fa_delta_kernel(const bf16* __restrict__ dout,
   float v = df.x * of.x + df.y * of.y;
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
-  if (lane == 0) delta[row] = v;
+  if (lane == 0) delta[didx] = v;
 }
 
 void launch_fa_delta(const void* dout, const void* o, float* delta,
-                     int64_t rows, hipStream_t stream) {
+                     int64_t rows, int Hq, int S, bool bshd,
+                     hipStream_t stream) {
   const int waves_per_wg = 8;
   const int64_t nwg = (rows + waves_per_wg - 1) / waves_per_wg;
   hipLaunchKernelGGL(fa_delta_kernel, dim3((uint32_t)nwg), dim3(waves_per_wg * 64),
-                     0, stream, (const bf16*)dout, (const bf16*)o, delta, rows);
+                     0, stream, (const bf16*)dout, (const bf16*)o, delta, rows,
+                     Hq, S, bshd ? 1 : 0);
 }
 
 // ---- launchers ------------------------------------------------------------
@@ -466,16 +487,16 @@ void launch_fa_delta(const void* dout, const void* o, float* delta,
 void launch_fa_bwd(const void* q, const void* k, const void* v, const void* dout,
                    const float* lse, const float* delta, void* dq, void* dk,
                    void* dv, int B, int Hq, int Hkv, int S, float scale,
-                   bool causal, hipStream_t stream) {
+                   bool causal, bool bshd, hipStream_t stream) {
   const int nblk = (S + FA_BLK * FA_WAVES - 1) / (FA_BLK * FA_WAVES);
   hipLaunchKernelGGL(fa_bwd_dkdv_kernel, dim3(nblk, B * Hkv), dim3(FA_THREADS), 0,
                      stream, (const bf16*)q, (const bf16*)k, (const bf16*)v,
                      (const bf16*)dout, lse, delta, (bf16*)dk, (bf16*)dv, B, Hq,
-                     Hkv, S, scale, causal ? 1 : 0);
+                     Hkv, S, scale, causal ? 1 : 0, bshd ? 1 : 0);
   hipLaunchKernelGGL(fa_bwd_dq_kernel, dim3(nblk, B * Hq), dim3(FA_THREADS), 0,
                      stream, (const bf16*)q, (const bf16*)k, (const bf16*)v,
                      (const bf16*)dout, lse, delta, (bf16*)dq, B, Hq, Hkv, S,
-                     scale, causal ? 1 : 0);
+                     scale, causal ? 1 : 0, bshd ? 1 : 0);
 }
 
 }  // namespace torchft_amd

@@ -78,9 +78,9 @@ struct KvRegs {
 };
 
 __device__ inline KvRegs load_kv(const bf16* __restrict__ ksrc,
-                                 const bf16* __restrict__ vsrc) {
+                                 const bf16* __restrict__ vsrc, int64_t ld) {
   const int t = threadIdx.x;
-  const int64_t off = (int64_t)(t >> 3) * FWD_D + ((t & 7) * 2) * 8;
+  const int64_t off = (int64_t)(t >> 3) * ld + ((t & 7) * 2) * 8;
   KvRegs r;
   r.k0 = reinterpret_cast<const uint4*>(ksrc + off)[0];
   r.k1 = reinterpret_cast<const uint4*>(ksrc + off)[1];
@@ -120,6 +120,13 @@ __device__ inline unsigned fwd_tr_lane_off(int lane) {
   asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(dst) : "v"(addr))
 #define FTR_WAIT2(r0, r1) \
   asm volatile("s_waitcnt lgkmcnt(0)" : "+v"(r0), "+v"(r1))
+
+// sequence-row offset of head (b,h): bshd=1 = [B,S,H,D] storage (see
+// flash_attn_bwd.hip row_off)
+__device__ inline int64_t frow_off(int b, int h, int s, int H, int S, int bshd) {
+  return bshd ? (((int64_t)b * S + s) * H + h) * FWD_D
+              : (((int64_t)b * H + h) * S + s) * FWD_D;
+}
 
 __device__ inline bf16x8_vec tr_join8(unsigned long long lo,
                                       unsigned long long hi) {
@@ -168,7 +175,7 @@ __global__ __launch_bounds__(FWD_THREADS, 1) void fa_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, bf16* __restrict__ out,
     float* __restrict__ lse, int B, int Hq, int Hkv, int S, float scale,
-    int causal) {
+    int causal, int bshd) {
   __shared__ SmemFwd sm;
   const int G = Hq / Hkv;
   const int b = blockIdx.y / Hq;
@@ -185,10 +192,10 @@ __global__ __launch_bounds__(FWD_THREADS, 1) void fa_fwd_kernel(
   const bool active = my_q0 < S;
 
   // Q block in registers: lane holds Q[q_glob][tt*16 + half*8 .. +8]
-  const int64_t q_head = ((int64_t)b * Hq + hq) * S;
+  const int64_t lse_head = ((int64_t)b * Hq + hq) * S;  // lse is dense BHS
   bf16x8_vec qfrag[8];
   if (active) {
-    const int64_t q_off = (q_head + q_glob) * FWD_D;
+    const int64_t q_off = frow_off(b, hq, q_glob, Hq, S, bshd);
 #pragma unroll
     for (int tt = 0; tt < 8; tt++) {
       qfrag[tt] =
@@ -196,9 +203,9 @@ __global__ __launch_bounds__(FWD_THREADS, 1) void fa_fwd_kernel(
     }
   }
 
-  const int64_t kv_head = ((int64_t)b * Hkv + hkv) * S;
-  const bf16* k_base = k + kv_head * FWD_D;
-  const bf16* v_base = v + kv_head * FWD_D;
+  const bf16* k_base = k + frow_off(b, hkv, 0, Hkv, S, bshd);
+  const bf16* v_base = v + frow_off(b, hkv, 0, Hkv, S, bshd);
+  const int64_t kv_ld = bshd ? (int64_t)Hkv * FWD_D : FWD_D;
 
   // number of KV tiles this WORKGROUP must stage
   const int kv_hi = causal ? min(S, Q0 + FWD_WAVES * FWD_QBLK) : S;
@@ -221,7 +228,7 @@ __global__ __launch_bounds__(FWD_THREADS, 1) void fa_fwd_kernel(
   const unsigned tr_off = fwd_tr_lane_off(lane);
   float* bc = sm.bcast[wave];
 
-  KvRegs staged = load_kv(k_base, v_base);  // tile 0
+  KvRegs staged = load_kv(k_base, v_base, kv_ld);  // tile 0
   for (int j = 0; j < nT; j++) {
     const int cur = j & 1;
     write_kv(&sm.k_img[cur], &sm.v_img[cur], staged);
@@ -230,8 +237,8 @@ __global__ __launch_bounds__(FWD_THREADS, 1) void fa_fwd_kernel(
       // issue the next tile's global loads now — they retire under this
       // tile's MFMA phase and are waited for by the write after the
       // next barrier (compiler-counted vmcnt)
-      staged = load_kv(k_base + (int64_t)(j + 1) * FWD_KVBLK * FWD_D,
-                       v_base + (int64_t)(j + 1) * FWD_KVBLK * FWD_D);
+      staged = load_kv(k_base + (int64_t)(j + 1) * FWD_KVBLK * kv_ld,
+                       v_base + (int64_t)(j + 1) * FWD_KVBLK * kv_ld, kv_ld);
     }
 
     const int key0 = j * FWD_KVBLK;
@@ -349,31 +356,31 @@ __global__ __launch_bounds__(FWD_THREADS, 1) void fa_fwd_kernel(
   // inactive waves returned above).
   if (half == 0) bc[l31] = 1.f / l_run;
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  const int64_t o_off = (q_head + my_q0) * FWD_D;
 #pragma unroll
   for (int dt = 0; dt < 4; dt++) {
 #pragma unroll
     for (int r = 0; r < 16; r++) {
       const int row = c_row(r, half);
-      out[o_off + (int64_t)row * FWD_D + dt * 32 + l31] =
+      out[frow_off(b, hq, my_q0 + row, Hq, S, bshd) + dt * 32 + l31] =
           __float2bfloat16(o_acc[dt][r] * bc[row]);
     }
   }
   if (half == 0) {
     // convert the log2-domain state back to the natural-log lse the
     // backward consumes: lse = ln2 * (m2 + log2(l))
-    lse[q_head + q_glob] = 0.6931471805599453f * (m_run + __log2f(l_run));
+    lse[lse_head + q_glob] = 0.6931471805599453f * (m_run + __log2f(l_run));
   }
 }
 
 void launch_fa_fwd(const void* q, const void* k, const void* v, void* out,
                    float* lse, int B, int Hq, int Hkv, int S, float scale,
-                   bool causal, hipStream_t stream) {
+                   bool causal, bool bshd, hipStream_t stream) {
   const int rows_per_wg = FWD_WAVES * FWD_QBLK;
   const int nblk = (S + rows_per_wg - 1) / rows_per_wg;
   hipLaunchKernelGGL(fa_fwd_kernel, dim3(nblk, B * Hq), dim3(FWD_THREADS), 0,
                      stream, (const bf16*)q, (const bf16*)k, (const bf16*)v,
-                     (bf16*)out, lse, B, Hq, Hkv, S, scale, causal ? 1 : 0);
+                     (bf16*)out, lse, B, Hq, Hkv, S, scale, causal ? 1 : 0,
+                     bshd ? 1 : 0);
 }
 
 }  // namespace torchft_amd

@@ -63,13 +63,13 @@ namespace torchft_amd {
 // flash_attn_bwd.hip --------------------------------------------------------
 void launch_fa_fwd(const void* q, const void* k, const void* v, void* out,
                    float* lse, int B, int Hq, int Hkv, int S, float scale,
-                   bool causal, tft_stream stream);
+                   bool causal, bool bshd, tft_stream stream);
 void launch_fa_delta(const void* dout, const void* o, float* delta,
-                     int64_t rows, tft_stream stream);
+                     int64_t rows, int Hq, int S, bool bshd, tft_stream stream);
 void launch_fa_bwd(const void* q, const void* k, const void* v, const void* dout,
                    const float* lse, const float* delta, void* dq, void* dk,
                    void* dv, int B, int Hq, int Hkv, int S, float scale,
-                   bool causal, tft_stream stream);
+                   bool causal, bool bshd, tft_stream stream);
 
 // mfma_probe.hip ------------------------------------------------------------
 void launch_mfma_probe(const void* A, const void* B, float* D, tft_stream s);

@@ -63,8 +63,9 @@ class _FlashAttentionFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dout):
         q, k, v, out, lse = ctx.saved_tensors
-        # D_i = rowsum(dout * out), fused fp32 kernel (no fp32 temporaries)
-        delta = hip_ext().fa_delta(dout.contiguous(), out)
+        # D_i = rowsum(dout * out), fused fp32 kernel (no fp32 temporaries;
+        # handles the model's [B,S,H,D]-backed transpose views natively)
+        delta = hip_ext().fa_delta(dout, out)
         dq, dk, dv = hip_ext().fa_bwd(
             q, k, v, dout, lse, delta, ctx.scale, ctx.causal
         )
@@ -96,9 +97,9 @@ def flash_attention(
     if scale is None:
         scale = q.shape[-1] ** -0.5
     if _supported(q, k):
-        return _FlashAttentionFn.apply(
-            q.contiguous(), k.contiguous(), v.contiguous(), causal, scale
-        )
+        # no .contiguous(): the kernels consume both the packed [B,H,S,D]
+        # layout and the model's [B,S,H,D]-backed transpose views natively
+        return _FlashAttentionFn.apply(q, k, v, causal, scale)
     return F.scaled_dot_product_attention(
         q, k, v, is_causal=causal, scale=scale, enable_gqa=q.shape[1] != k.shape[1]
     )

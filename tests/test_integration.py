@@ -142,7 +142,13 @@ def assert_state_dicts_equal(dicts: List[Dict[str, torch.Tensor]]) -> None:
 
 class TestFTIntegration:
     def test_healthy_two_replicas(self):
-        dicts = _run_replicas(2, total_steps=5, injector=EventInjector())
+        # min_replicas=2: under heavy suite load a heartbeat flap can let a
+        # replica form a lone quorum and commit an un-averaged step — legal
+        # protocol behavior (DYNAMIC world), but this test asserts the
+        # lockstep trajectory, so require both members in every quorum
+        dicts = _run_replicas(
+            2, total_steps=5, injector=EventInjector(), min_replicas=2
+        )
         assert_state_dicts_equal(dicts)
 
     def test_replica_failure_and_recovery(self):

@@ -240,6 +240,20 @@ class TestLlamaSmokeGPU:
         assert torch.isfinite(loss)
 
 
+
+def _q8_roundtrip(x: torch.Tensor) -> torch.Tensor:
+    """Block fp8 quantize->dequantize (the wire format's information loss)."""
+    from torchft_amd.quantization import FP8_MAX, QBLOCK
+
+    nb = (x.numel() + QBLOCK - 1) // QBLOCK
+    xp = torch.zeros(nb * QBLOCK, device=x.device)
+    xp[: x.numel()] = x.float()
+    xb = xp.view(nb, QBLOCK)
+    amax = xb.abs().amax(1, keepdim=True).clamp_min(1e-30)
+    q = (xb * (FP8_MAX / amax)).to(torch.float8_e4m3fn).float() * (amax / FP8_MAX)
+    return q.view(-1)[: x.numel()]
+
+
 class _MailboxCtx:
     """Shared exchange state for N virtual ranks on one device."""
 
@@ -331,7 +345,13 @@ class TestQuantizedAllreduceMultiRank:
         inputs = [
             torch.randn(n, device=dev, dtype=torch.bfloat16) for _ in range(world)
         ]
-        expected = torch.stack([t.float() for t in inputs]).sum(0)
+        # the exact pipeline loses information twice (quantize inputs,
+        # requantize the reduced slice): compare against the same
+        # double-quantized reference, not the exact sum — the exact-sum
+        # residual legitimately reaches ~0.35 abs at 100k N(0,1) elements
+        expected = _q8_roundtrip(
+            torch.stack([_q8_roundtrip(t) for t in inputs]).sum(0)
+        )
         if op_avg:
             expected /= world
         ctx = _MailboxCtx(world)
@@ -349,10 +369,9 @@ class TestQuantizedAllreduceMultiRank:
         with ThreadPoolExecutor(max_workers=world) as ex:
             outs = list(ex.map(rank_main, range(world)))
 
-        tol = 0.15 if not op_avg else 0.15 / world * 2
         for r, out in enumerate(outs):
             torch.testing.assert_close(
-                out.float(), expected, rtol=0.15, atol=max(tol, 0.05),
+                out.float(), expected, rtol=0.06, atol=0.06,
                 msg=f"rank {r} mismatch at world {world}",
             )
 
@@ -370,7 +389,9 @@ class TestQuantizedAllreduceMultiRank:
         inputs = [
             torch.randn(n, device=dev, dtype=torch.bfloat16) for _ in range(world)
         ]
-        total = torch.stack([t.float() for t in inputs]).sum(0)
+        total = _q8_roundtrip(
+            torch.stack([_q8_roundtrip(t) for t in inputs]).sum(0)
+        )
         _, _, bpr, _ = Q.pack_geometry([inputs[0]], world)
         ctx = _MailboxCtx(world)
 
@@ -392,7 +413,7 @@ class TestQuantizedAllreduceMultiRank:
             if lo >= n:
                 continue
             torch.testing.assert_close(
-                out[: hi - lo].float(), total[lo:hi], rtol=0.15, atol=0.15,
+                out[: hi - lo].float(), total[lo:hi], rtol=0.06, atol=0.06,
                 msg=f"rank {r} slice mismatch at world {world}",
             )
 

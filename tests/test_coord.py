@@ -6,6 +6,7 @@ Mirrors the reference's Rust unit tests: quorum_compute pure-function cases
 """
 
 import threading
+import time
 from datetime import timedelta
 
 import pytest
@@ -349,3 +350,50 @@ class TestResilienceCoord:
         finally:
             mgr.shutdown()
             lh.shutdown()
+
+
+class TestLighthouseFailover:
+    def test_manager_survives_lighthouse_restart(self):
+        """Kill the lighthouse and bring a new one up at the SAME address:
+        the manager's retry/backoff client must reconnect and the next
+        quorum succeed (managers are lighthouse-restart tolerant)."""
+        import socket
+
+        # reserve a port so the replacement can bind the same address
+        probe = socket.socket()
+        probe.bind(("127.0.0.1", 0))
+        port = probe.getsockname()[1]
+        probe.close()
+
+        lh = core.LighthouseServer(
+            bind=f"127.0.0.1:{port}", min_replicas=1, join_timeout_ms=100
+        )
+        mgr = core.ManagerServer(
+            replica_id="fo0",
+            lighthouse_addr=lh.address(),
+            hostname="127.0.0.1",
+            bind="127.0.0.1:0",
+            store_addr="s",
+            world_size=1,
+            heartbeat_interval=TD(milliseconds=50),
+            connect_timeout=TD(seconds=5),
+            quorum_retries=3,
+        )
+        lh2 = None
+        try:
+            c = core.ManagerClient(mgr.address(), connect_timeout=TD(seconds=5))
+            q1 = c._quorum(0, 0, "m", False, TD(seconds=10))
+            assert q1.quorum_id >= 0
+
+            lh.shutdown()
+            time.sleep(0.3)
+            lh2 = core.LighthouseServer(
+                bind=f"127.0.0.1:{port}", min_replicas=1, join_timeout_ms=100
+            )
+            q2 = c._quorum(0, 1, "m", False, TD(seconds=15))
+            assert q2.quorum_id >= 0
+            assert q2.replica_ids == ["fo0"]
+        finally:
+            mgr.shutdown()
+            if lh2 is not None:
+                lh2.shutdown()

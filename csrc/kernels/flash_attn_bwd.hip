@@ -280,8 +280,11 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dkdv_kernel(
       // + vmcnt(0) drain inside this loop)
 #pragma unroll
       for (int it = 0; it < 8; it++) {
-        const int dt = it >> 1;
-        const int h2 = it & 1;
+        // dt fastest: consecutive MFMAs rotate over the four dv/dk
+        // accumulators, so the 32-cycle dependent-accumulator latency of
+        // the 32x32 MFMA never stalls the pipe (PMC: SQ_WAIT_INST_ANY)
+        const int dt = it & 3;
+        const int h2 = it >> 2;
         const unsigned base = dt * (2 * SUBT) + h2 * 512;
         unsigned long long fd0, fd1, fq0, fq1;
         TR_READ(fd0, b_base + base);        // dO rows +0..3
@@ -407,13 +410,14 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
       TR_READ(fk1[0], a_base + 128);
 #pragma unroll
       for (int it = 0; it < 8; it++) {
-        const int dt = it >> 1;
-        const int h2 = it & 1;
+        // dt fastest (see dkdv): rotate the four dq accumulators
+        const int dt = it & 3;
+        const int h2 = it >> 2;
         const int cur = it & 1;
         const int nxt = cur ^ 1;
         if (it < 7) {
           const int it2 = it + 1;
-          const unsigned nbase = (it2 >> 1) * (2 * SUBT) + (it2 & 1) * 512;
+          const unsigned nbase = (it2 & 3) * (2 * SUBT) + (it2 >> 2) * 512;
           TR_READ(fk0[nxt], a_base + nbase);
           TR_READ(fk1[nxt], a_base + nbase + 128);
           TR_WAIT2_KEEP(2, fk0[cur], fk1[cur]);

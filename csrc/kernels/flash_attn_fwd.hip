@@ -328,20 +328,21 @@ __global__ __launch_bounds__(FWD_THREADS, 1) void fa_fwd_kernel(
         pa1.u[2] = pk[6];
         pa1.u[3] = pk[7];
 
-        // PV: V B-fragments by hardware transpose; k-dim = 16 keys per h2
+        // PV: V B-fragments by hardware transpose; k-dim = 16 keys per
+        // h2. dt fastest so consecutive MFMAs rotate the four O
+        // accumulators (dependent-latency hiding; see bwd kernels).
 #pragma unroll
-        for (int dt = 0; dt < 4; dt++) {
-#pragma unroll
-          for (int h2 = 0; h2 < 2; h2++) {
-            const unsigned base =
-                v_img_base + dt * (2 * FSUBT) + (s * 32 + h2 * 16) * 32;
-            unsigned long long t0, t1;
-            FTR_READ(t0, base);
-            FTR_READ(t1, base + 128);
-            FTR_WAIT2(t0, t1);
-            o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                h2 ? pa1.v : pa0.v, tr_join8(t0, t1), o_acc[dt], 0, 0, 0);
-          }
+        for (int it = 0; it < 8; it++) {
+          const int dt = it & 3;
+          const int h2 = it >> 2;
+          const unsigned base =
+              v_img_base + dt * (2 * FSUBT) + (s * 32 + h2 * 16) * 32;
+          unsigned long long t0, t1;
+          FTR_READ(t0, base);
+          FTR_READ(t1, base + 128);
+          FTR_WAIT2(t0, t1);
+          o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              h2 ? pa1.v : pa0.v, tr_join8(t0, t1), o_acc[dt], 0, 0, 0);
         }
       }
     }

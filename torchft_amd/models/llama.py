@@ -202,8 +202,11 @@ class Llama(nn.Module):
         losses = []
         for i in range(0, n, chunk_rows):
             logits = self.output(h[i : i + chunk_rows])
+            # no .float(): torch's CE softmax accumulates in fp32 for bf16
+            # inputs anyway, and the explicit cast materialized a 2 GB fp32
+            # logits copy per chunk (plus its gradient) at the 128k vocab
             losses.append(
-                F.cross_entropy(logits.float(), t[i : i + chunk_rows], reduction="sum")
+                F.cross_entropy(logits, t[i : i + chunk_rows], reduction="sum").float()
             )
         return torch.stack(losses).sum() / n
 

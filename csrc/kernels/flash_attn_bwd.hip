@@ -319,6 +319,58 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dkdv_kernel(
   }
 }
 
+// ---- dq-kernel 64-key tile geometry ---------------------------------------
+// [8 d-subtiles][64 rows][16 cols] with a 16-B pad: staging a 64-key K/V
+// tile per barrier pair halves the per-MFMA staging/barrier overhead vs
+// the 32-key tiles the dkdv kernel uses for its Q/dO stream.
+#define SUBT64 2080
+
+struct Tile64 {
+  __align__(16) unsigned char sub[8 * SUBT64];
+};
+
+struct SmemDq {
+  Tile64 k_img[2];
+  Tile64 v_img[2];
+  __align__(16) unsigned char pb[FA_WAVES][32 * 64];
+};
+
+__device__ inline int st64_addr(int tt, int row, int byte_in_row) {
+  return tt * SUBT64 + row * 32 + byte_in_row;
+}
+
+// thread t of 512 stages row r=t>>3, 32-B chunk pair c0=(t&7)*2 of each
+// [64][128] tile (one uint4 pair per tensor)
+__device__ inline void write_tiles64(Tile64* kd, Tile64* vd,
+                                     const bf16* __restrict__ ksrc,
+                                     const bf16* __restrict__ vsrc,
+                                     int64_t ld) {
+  const int t = threadIdx.x;
+  const int r = t >> 3;
+  const int c0 = (t & 7) * 2;
+  const int64_t off = (int64_t)r * ld + c0 * 8;
+  const uint4 k0 = reinterpret_cast<const uint4*>(ksrc + off)[0];
+  const uint4 k1 = reinterpret_cast<const uint4*>(ksrc + off)[1];
+  const uint4 v0 = reinterpret_cast<const uint4*>(vsrc + off)[0];
+  const uint4 v1 = reinterpret_cast<const uint4*>(vsrc + off)[1];
+  const int a0 = st64_addr(c0 >> 1, r, (c0 & 1) * 16);
+  *reinterpret_cast<uint4*>(kd->sub + a0) = k0;
+  *reinterpret_cast<uint4*>(kd->sub + a0 + 16) = k1;
+  *reinterpret_cast<uint4*>(vd->sub + a0) = v0;
+  *reinterpret_cast<uint4*>(vd->sub + a0 + 16) = v1;
+}
+
+__device__ inline bf16x8_vec rm64_bfrag(const unsigned char* tile, int tt,
+                                        int row, int half) {
+  return *reinterpret_cast<const bf16x8_vec*>(
+      tile + st64_addr(tt, row, half * 16));
+}
+
+// tr_b16 lane invariant in the 64-row subtiles
+__device__ inline unsigned tr64_lane_off(int lane) {
+  return ((lane >> 4) & 1) * SUBT64 + ((lane >> 5) * 8) * 32 + (lane & 15) * 8;
+}
+
 // ---- kernel 2: dQ ---------------------------------------------------------
 __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
@@ -326,7 +378,7 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
     const float* __restrict__ lse, const float* __restrict__ delta,
     bf16* __restrict__ dq, int B, int Hq, int Hkv, int S, float scale,
     int causal, int bshd) {
-  __shared__ SmemFA sm;
+  __shared__ SmemDq sm;
   const int G = Hq / Hkv;
   const int b = blockIdx.y / Hq;
   const int hq = blockIdx.y % Hq;
@@ -337,8 +389,9 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
   const int l31 = lane & 31;
   const int ib = blockIdx.x * FA_WAVES + wave;
   const bool active = ib * FA_BLK < S;
-  const int nK = S / FA_BLK;
-  const int T = causal ? min(blockIdx.x * FA_WAVES + FA_WAVES, nK) : nK;
+  const int nK64 = (S + 63) / 64;
+  const int kv_hi = causal ? (blockIdx.x * FA_WAVES + FA_WAVES) * FA_BLK : S;
+  const int T = causal ? (min(kv_hi, S) + 63) / 64 : nK64;
 
   bf16x8_vec qfrag[8], dofrag[8];
   float lse_row[16], delta_row[16];
@@ -362,70 +415,78 @@ __global__ __launch_bounds__(FA_THREADS, 1) void fa_bwd_dq_kernel(
   f32x16 dq_acc[4] = {};
 
   const int64_t kv_ld = bshd ? (int64_t)Hkv * FA_D : FA_D;
-  const unsigned tr_off = tr_lane_off(lane);
+  const unsigned tr_off = tr64_lane_off(lane);
 
   for (int j = 0; j < T; j++) {
     const int cur = j & 1;
-    {
-      TileRegs r = load_tiles(k + row_off(b, hkv, j * FA_BLK, Hkv, S, bshd),
-                              v + row_off(b, hkv, j * FA_BLK, Hkv, S, bshd),
-                              kv_ld);
-      write_tiles(&sm.img[cur], r);
-    }
+    write_tiles64(&sm.k_img[cur], &sm.v_img[cur],
+                  k + row_off(b, hkv, j * 64, Hkv, S, bshd),
+                  v + row_off(b, hkv, j * 64, Hkv, S, bshd), kv_ld);
     __syncthreads();
 
-    if (active && !(causal && j > ib)) {
-      const ImageSet* img = &sm.img[cur];
-      f32x16 s_acc = {};
-      f32x16 dp_acc = {};
-#pragma unroll
-      for (int tt = 0; tt < 8; tt++) {
-        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            qfrag[tt], rm_bfrag(img->a.sub, tt, half, l31), s_acc, 0, 0, 0);
-        dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            dofrag[tt], rm_bfrag(img->b.sub, tt, half, l31), dp_acc, 0, 0, 0);
-      }
-
+    if (active && !(causal && j * 64 > ib * FA_BLK + FA_BLK - 1)) {
+      const unsigned char* kimg = sm.k_img[cur].sub;
+      const unsigned char* vimg = sm.v_img[cur].sub;
+      const unsigned a_base = (unsigned)(uintptr_t)kimg + tr_off;
       unsigned char* pb = sm.pb[wave];
-      const int kg = j * FA_BLK + l31;
-#pragma unroll
-      for (int rg = 0; rg < 16; rg++) {
-        const int qg = ib * FA_BLK + c_row(rg, half);
-        const bool valid = !causal || (qg >= kg);
-        const float p = valid ? __expf(scale * s_acc[rg] - lse_row[rg]) : 0.0f;
-        const float ds = p * (dp_acc[rg] - delta_row[rg]) * scale;
-        *reinterpret_cast<bf16*>(pb + pb_addr(c_row(rg, half), l31 * 2)) =
-            __float2bfloat16(ds);
-      }
 
-      bf16x8_vec pbf[2];
 #pragma unroll
-      for (int h2 = 0; h2 < 2; h2++) pbf[h2] = pb_afrag(pb, h2, half, l31);
+      for (int s = 0; s < 2; s++) {  // two 32-key halves per staged tile
+        const int kb0 = j * 64 + s * 32;
+        if (causal && kb0 > ib * FA_BLK + FA_BLK - 1) break;
 
-      // dQ += dS K — K B-fragments by hardware transpose, pipelined one
-      // iteration ahead (see dkdv kernel)
-      const unsigned a_base = (unsigned)(uintptr_t)img->a.sub + tr_off;
-      unsigned long long fk0[2], fk1[2];
-      TR_READ(fk0[0], a_base);
-      TR_READ(fk1[0], a_base + 128);
+        f32x16 s_acc = {};
+        f32x16 dp_acc = {};
 #pragma unroll
-      for (int it = 0; it < 8; it++) {
-        // dt fastest (see dkdv): rotate the four dq accumulators
-        const int dt = it & 3;
-        const int h2 = it >> 2;
-        const int cur = it & 1;
-        const int nxt = cur ^ 1;
-        if (it < 7) {
-          const int it2 = it + 1;
-          const unsigned nbase = (it2 & 3) * (2 * SUBT) + (it2 >> 2) * 512;
-          TR_READ(fk0[nxt], a_base + nbase);
-          TR_READ(fk1[nxt], a_base + nbase + 128);
-          TR_WAIT2_KEEP(2, fk0[cur], fk1[cur]);
-        } else {
-          TR_WAIT2_KEEP(0, fk0[cur], fk1[cur]);
+        for (int tt = 0; tt < 8; tt++) {
+          s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              qfrag[tt], rm64_bfrag(kimg, tt, s * 32 + l31, half), s_acc, 0, 0,
+              0);
+          dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              dofrag[tt], rm64_bfrag(vimg, tt, s * 32 + l31, half), dp_acc, 0,
+              0, 0);
         }
-        dq_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            pbf[h2], tr_join(fk0[cur], fk1[cur]), dq_acc[dt], 0, 0, 0);
+
+        const int kg = kb0 + l31;
+#pragma unroll
+        for (int rg = 0; rg < 16; rg++) {
+          const int qg = ib * FA_BLK + c_row(rg, half);
+          const bool valid = !causal || (qg >= kg);
+          const float p = valid ? __expf(scale * s_acc[rg] - lse_row[rg]) : 0.0f;
+          const float ds = p * (dp_acc[rg] - delta_row[rg]) * scale;
+          *reinterpret_cast<bf16*>(pb + pb_addr(c_row(rg, half), l31 * 2)) =
+              __float2bfloat16(ds);
+        }
+
+        bf16x8_vec pbf[2];
+#pragma unroll
+        for (int h2 = 0; h2 < 2; h2++) pbf[h2] = pb_afrag(pb, h2, half, l31);
+
+        // dQ += dS K — K B-fragments by hardware transpose, pipelined one
+        // iteration ahead; dt fastest rotates the four accumulators
+        const unsigned s_base = a_base + (s * 32) * 32;
+        unsigned long long fk0[2], fk1[2];
+        TR_READ(fk0[0], s_base);
+        TR_READ(fk1[0], s_base + 128);
+#pragma unroll
+        for (int it = 0; it < 8; it++) {
+          const int dt = it & 3;
+          const int h2 = it >> 2;
+          const int cur2 = it & 1;
+          const int nxt = cur2 ^ 1;
+          if (it < 7) {
+            const int it2 = it + 1;
+            const unsigned nbase =
+                (it2 & 3) * (2 * SUBT64) + (it2 >> 2) * 512;
+            TR_READ(fk0[nxt], s_base + nbase);
+            TR_READ(fk1[nxt], s_base + nbase + 128);
+            TR_WAIT2_KEEP(2, fk0[cur2], fk1[cur2]);
+          } else {
+            TR_WAIT2_KEEP(0, fk0[cur2], fk1[cur2]);
+          }
+          dq_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              pbf[h2], tr_join(fk0[cur2], fk1[cur2]), dq_acc[dt], 0, 0, 0);
+        }
       }
     }
     __syncthreads();

@@ -110,3 +110,24 @@ class TestFusedAdamWStepBuckets:
                 opt.zero_grad(set_to_none=True)
         torch.testing.assert_close(p_fused[0], p_ref[0], rtol=1e-5, atol=1e-6)
         torch.testing.assert_close(p_fused[1], p_ref[1], rtol=1e-5, atol=1e-6)
+
+
+class TestLlama70BConfig:
+    def test_70b_parameter_count(self):
+        """The 70B config must actually be Llama-3-70B-sized (the 8-GPU HSDP
+        bench path instantiates it; meta device keeps this test cheap)."""
+        from torchft_amd.models.llama import LLAMA3_70B, Llama
+
+        with torch.device("meta"):
+            m = Llama(LLAMA3_70B, dtype=torch.bfloat16,
+                      checkpoint_activations=True)
+        n = m.num_params()
+        assert 68e9 < n < 73e9, f"70B config has {n/1e9:.1f}B params"
+
+    def test_8b_parameter_count(self):
+        from torchft_amd.models.llama import LLAMA3_8B, Llama
+
+        with torch.device("meta"):
+            m = Llama(LLAMA3_8B, dtype=torch.bfloat16)
+        n = m.num_params()
+        assert 7.5e9 < n < 8.5e9, f"8B config has {n/1e9:.1f}B params"

@@ -142,11 +142,13 @@ class _SerialExecutor:
         )
         self._thread = threading.Thread(target=self._run, name=name, daemon=True)
         self._started = False
+        self._start_mu = threading.Lock()
 
     def submit(self, fn: Callable[[], None]) -> _Pending:
-        if not self._started:
-            self._thread.start()
-            self._started = True
+        with self._start_mu:
+            if not self._started:
+                self._thread.start()
+                self._started = True
         pending = _Pending()
         self._q.put((fn, pending))
         return pending

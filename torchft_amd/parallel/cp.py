@@ -471,3 +471,204 @@ def ring_attention(
         pg, rank, world, causal,
     )
     return out.transpose(1, 2)
+
+
+# ---------------------------------------------------------------------------
+# zigzag ring attention (balanced causal ring)
+# ---------------------------------------------------------------------------
+
+
+def shard_sequence_zigzag(
+    t: torch.Tensor, rank: int, world: int, dim: int = 1
+) -> torch.Tensor:
+    """Zigzag shard: the sequence splits into 2W chunks and rank i owns
+    chunks (i, 2W-1-i) concatenated — every rank then sees the same causal
+    work per ring hop, fixing the plain ring's tail-heavy imbalance."""
+    seq = t.size(dim)
+    assert seq % (2 * world) == 0, (
+        f"sequence {seq} not divisible by 2*world {2 * world}"
+    )
+    c = seq // (2 * world)
+    lo = t.narrow(dim, rank * c, c)
+    hi = t.narrow(dim, (2 * world - 1 - rank) * c, c)
+    return torch.cat([lo, hi], dim=dim).contiguous()
+
+
+def unshard_sequence_zigzag(
+    parts: list, world: int, dim: int = 1
+) -> torch.Tensor:
+    """Inverse of :func:`shard_sequence_zigzag` given every rank's shard."""
+    chunks = [None] * (2 * world)
+    for r, p in enumerate(parts):
+        c = p.size(dim) // 2
+        chunks[r] = p.narrow(dim, 0, c)
+        chunks[2 * world - 1 - r] = p.narrow(dim, c, c)
+    return torch.cat(chunks, dim=dim)
+
+
+class _ZigzagRingAttention(torch.autograd.Function):
+    """Causal ring attention over zigzag shards.
+
+    Each rank's Q/K/V shard is [low chunk i | high chunk 2W-1-i]. Per hop
+    the owner's two KV chunks are tested against the rank's two Q chunks —
+    chunk-causal visibility (kv chunk <= q chunk; equality = in-chunk
+    causal) gives every rank the same number of visible blocks per hop.
+    Forward streams KV around the ring with per-q-chunk online LSE merges;
+    backward re-all-gathers KV and re-enters flash backward per visible
+    block with the merged out/lse (same identity as _RingAttention).
+    """
+
+    @staticmethod
+    def forward(ctx, q, k_shard, v_shard, pg, rank, world):
+        B, Hq, s2, D = q.shape  # s2 = 2c
+        c = s2 // 2
+        q = q.contiguous()
+        Hkv = k_shard.shape[1]
+        rep = Hq // Hkv
+        if rep > 1 and _native_gqa_ok(q.device.type, q.dtype):
+            rep = 1
+
+        my_chunks = (rank, 2 * world - 1 - rank)
+        q_halves = (q[:, :, :c], q[:, :, c:])
+        out_acc = [
+            torch.zeros(B, Hq, c, D, dtype=torch.float32, device=q.device)
+            for _ in range(2)
+        ]
+        lse_acc = [
+            torch.full((B, Hq, c), float("-inf"), dtype=torch.float32,
+                       device=q.device)
+            for _ in range(2)
+        ]
+
+        cur_k, cur_v = k_shard.contiguous(), v_shard.contiguous()
+        nxt, prv = (rank + 1) % world, (rank - 1) % world
+        aux_by_block = {}
+        for hop in range(world):
+            owner = (rank - hop) % world
+            if hop < world - 1:
+                send_buf = torch.cat(
+                    [cur_k.reshape(1, -1), cur_v.reshape(1, -1)], 0
+                )
+                recv_buf = torch.empty_like(send_buf)
+                if rank % 2 == 0:
+                    ws = pg.send([send_buf], nxt, tag=hop)
+                    wr = pg.recv([recv_buf], prv, tag=hop)
+                else:
+                    wr = pg.recv([recv_buf], prv, tag=hop)
+                    ws = pg.send([send_buf], nxt, tag=hop)
+                ws.wait()
+                wr.wait()
+
+            owner_chunks = (owner, 2 * world - 1 - owner)
+            for kh in range(2):
+                kc = owner_chunks[kh]
+                kb = _expand_kv(cur_k[:, :, kh * c : (kh + 1) * c].contiguous(), rep)
+                vb = _expand_kv(cur_v[:, :, kh * c : (kh + 1) * c].contiguous(), rep)
+                for qh in range(2):
+                    qc = my_chunks[qh]
+                    if kc > qc:
+                        continue
+                    blk_causal = kc == qc
+                    o_b, l_b, aux = _flash_fwd_raw(
+                        q_halves[qh].contiguous(), kb, vb, causal=blk_causal
+                    )
+                    aux_by_block[(qh, kc)] = aux
+                    out_acc[qh], lse_acc[qh] = _merge_block(
+                        out_acc[qh], lse_acc[qh], o_b, l_b
+                    )
+            if hop < world - 1:
+                cur_k = recv_buf[0].view_as(cur_k).contiguous()
+                cur_v = recv_buf[1].view_as(cur_v).contiguous()
+
+        out = torch.cat([a.to(q.dtype) for a in out_acc], dim=2)
+        ctx.save_for_backward(q, k_shard, v_shard, out, lse_acc[0], lse_acc[1])
+        ctx.pg, ctx.rank, ctx.world, ctx.rep = pg, rank, world, rep
+        ctx.aux_by_block = aux_by_block
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k_shard, v_shard, out, lse0, lse1 = ctx.saved_tensors
+        pg, rank, world, rep = ctx.pg, ctx.rank, ctx.world, ctx.rep
+        B, Hkv, s2, D = k_shard.shape
+        c = s2 // 2
+        Hq = q.shape[1]
+        dout = dout.contiguous()
+
+        flat = torch.cat([k_shard.reshape(1, -1), v_shard.reshape(1, -1)], 0)
+        gath = torch.empty(world * flat.numel(), dtype=flat.dtype,
+                           device=flat.device)
+        pg.allgather_into_tensor_coalesced(
+            [gath], [flat.reshape(-1)], AllgatherOptions()
+        ).wait()
+        gath = gath.view(world, 2, B, Hkv, 2, c, D)  # [owner][k/v][..][half]
+
+        my_chunks = (rank, 2 * world - 1 - rank)
+        q_halves = (q[:, :, :c].contiguous(), q[:, :, c:].contiguous())
+        o_halves = (out[:, :, :c].contiguous(), out[:, :, c:].contiguous())
+        d_halves = (dout[:, :, :c].contiguous(), dout[:, :, c:].contiguous())
+        lses = (lse0, lse1)
+
+        dq = torch.zeros_like(q, dtype=torch.float32)
+        dkv_chunks = torch.zeros(2, 2 * world, B, Hkv, c, D,
+                                 dtype=torch.float32, device=q.device)
+
+        def fold(d):
+            if rep == 1:
+                return d.float()
+            return d.view(B, Hkv, rep, c, D).float().sum(2)
+
+        for o in range(world):
+            owner_chunks = (o, 2 * world - 1 - o)
+            for kh in range(2):
+                kc = owner_chunks[kh]
+                kb = _expand_kv(
+                    gath[o, 0, :, :, kh].to(q.dtype).contiguous(), rep
+                )
+                vb = _expand_kv(
+                    gath[o, 1, :, :, kh].to(q.dtype).contiguous(), rep
+                )
+                for qh in range(2):
+                    qc = my_chunks[qh]
+                    if kc > qc:
+                        continue
+                    dq_b, dk_b, dv_b = _flash_bwd_raw(
+                        d_halves[qh], q_halves[qh], kb, vb, o_halves[qh],
+                        lses[qh], kc == qc, ctx.aux_by_block[(qh, kc)]
+                    )
+                    dq[:, :, qh * c : (qh + 1) * c] += dq_b.float()
+                    dkv_chunks[0, kc] += fold(dk_b)
+                    dkv_chunks[1, kc] += fold(dv_b)
+
+        opts = AllreduceOptions()
+        opts.reduceOp = ReduceOp.SUM
+        pg.allreduce([dkv_chunks.view(-1)], opts).wait()
+        lo, hi = my_chunks
+        dk = torch.cat([dkv_chunks[0, lo], dkv_chunks[0, hi]], dim=2).to(
+            k_shard.dtype
+        )
+        dv = torch.cat([dkv_chunks[1, lo], dkv_chunks[1, hi]], dim=2).to(
+            v_shard.dtype
+        )
+        return dq.to(q.dtype), dk, dv, None, None, None
+
+
+def ring_attention_zigzag(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    pg: Optional[ProcessGroup],
+    rank: int,
+    world: int,
+) -> torch.Tensor:
+    """Balanced causal ring attention over :func:`shard_sequence_zigzag`
+    shards ([B, 2c, H, D] in/out)."""
+    if pg is None or world == 1:
+        return cp_attention(q, k, v, None, rank, 1, True)
+    out = _ZigzagRingAttention.apply(
+        q.transpose(1, 2).contiguous(),
+        k.transpose(1, 2).contiguous(),
+        v.transpose(1, 2).contiguous(),
+        pg, rank, world,
+    )
+    return out.transpose(1, 2)

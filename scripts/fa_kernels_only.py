@@ -16,5 +16,6 @@ out, lse, *_ = torch.ops.aten._scaled_dot_product_flash_attention(
 delta = hip_ext().fa_delta(dout, out)
 for _ in range(3):
     hip_ext().fa_bwd(q, k, v, dout, lse, delta, D ** -0.5, True)
+    hip_ext().fa_fwd(q, k, v, D ** -0.5, True)
 torch.cuda.synchronize()
 print("done")

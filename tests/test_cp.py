@@ -49,7 +49,9 @@ class TestContextParallel:
             qs = shard_sequence(q, rank, world)
             ks = shard_sequence(k, rank, world)
             vs = shard_sequence(v, rank, world)
-            return cp_attention(qs, ks, vs, pg, rank, world, causal)
+            out = cp_attention(qs, ks, vs, pg, rank, world, causal)
+            pg.shutdown()
+            return out
 
         with ThreadPoolExecutor(max_workers=world) as ex:
             outs = list(ex.map(worker, range(world)))
@@ -182,7 +184,9 @@ class TestLlamaCP:
             m.eval()
             shard = shard_sequence(toks, rank, world, dim=1)
             with torch.no_grad():
-                return m.forward_hidden(shard)
+                out = m.forward_hidden(shard)
+            pg.shutdown()
+            return out
 
         with ThreadPoolExecutor(max_workers=world) as ex:
             outs = list(ex.map(worker, range(world)))
@@ -216,6 +220,7 @@ class TestRingAttention:
             vs = shard_sequence(v.detach(), rank, world).requires_grad_(True)
             out = ring_attention(qs, ks, vs, pg, rank, world, causal)
             out.backward(shard_sequence(g, rank, world))
+            pg.shutdown()
             return out.detach(), qs.grad, ks.grad, vs.grad
 
         with ThreadPoolExecutor(max_workers=world) as ex:
@@ -271,6 +276,7 @@ class TestZigzagRingAttention:
             vs = shard_sequence_zigzag(v.detach(), rank, world).requires_grad_(True)
             out = ring_attention_zigzag(qs, ks, vs, pg, rank, world)
             out.backward(shard_sequence_zigzag(g, rank, world))
+            pg.shutdown()
             return out.detach(), qs.grad, ks.grad, vs.grad
 
         with ThreadPoolExecutor(max_workers=world) as ex:

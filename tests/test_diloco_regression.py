@@ -94,7 +94,7 @@ def _run_replica(replica_id: int, lighthouse_addr: str, outer_steps: int) -> Lis
 
 class TestDiLoCoRegression:
     def test_trajectory_matches_fixture(self):
-        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=100)
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=1000)
         try:
             with ThreadPoolExecutor(max_workers=2) as ex:
                 futs = [ex.submit(_run_replica, i, lh.address(), 3) for i in range(2)]
@@ -294,7 +294,7 @@ class TestStreamingDiLoCoRegression:
     def test_delayed_sync_trajectory(self):
         """Streaming DiLoCo with fragment_sync_delay=1: the allreduce is
         staged one inner step before it commits."""
-        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=100)
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=1000)
         try:
             with ThreadPoolExecutor(max_workers=2) as ex:
                 futs = [
@@ -309,7 +309,7 @@ class TestStreamingDiLoCoRegression:
     def test_commit_failure_recovery_trajectory(self):
         """An injected allreduce failure must roll the window back; the
         post-recovery trajectory is pinned by the fixture."""
-        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=100)
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=1000)
         try:
             with ThreadPoolExecutor(max_workers=2) as ex:
                 futs = [
@@ -326,7 +326,7 @@ class TestStreamingDiLoCoRegression:
 
 class TestLocalSGDRegression:
     def test_localsgd_trajectory(self):
-        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=100)
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=1000)
         try:
             with ThreadPoolExecutor(max_workers=2) as ex:
                 futs = [

@@ -122,7 +122,7 @@ def _replica_main(
 def _run_replicas(
     num_replicas: int, total_steps: int, injector: EventInjector, min_replicas: int = 1
 ) -> List[Dict[str, torch.Tensor]]:
-    lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=min_replicas, join_timeout_ms=100)
+    lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=min_replicas, join_timeout_ms=1000)
     try:
         with ThreadPoolExecutor(max_workers=num_replicas) as ex:
             futs = [

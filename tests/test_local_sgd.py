@@ -69,7 +69,7 @@ def _localsgd_replica(replica_id: int, lighthouse_addr: str, total_outer: int,
 
 class TestLocalSGD:
     def test_two_replicas_converge(self):
-        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=100)
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=1000)
         try:
             with ThreadPoolExecutor(max_workers=2) as ex:
                 futs = [
@@ -159,7 +159,7 @@ def _diloco_replica(
 
 class TestDiLoCo:
     def test_two_replicas_healthy(self):
-        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=100)
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=1000)
         try:
             with ThreadPoolExecutor(max_workers=2) as ex:
                 futs = [
@@ -206,7 +206,7 @@ class TestDiLoCo:
             lh.shutdown()
 
     def test_recovery_after_failure(self):
-        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=100)
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=1000)
         try:
             with ThreadPoolExecutor(max_workers=2) as ex:
                 futs = [

@@ -44,14 +44,14 @@ typedef __attribute__((__vector_size__(16 * sizeof(float)))) float f32x16;
 
 #define FWD_D 128
 #define FWD_QBLK 32
-#define FWD_KVBLK 128
+#define FWD_KVBLK 64
 #define FWD_WAVES 8
 #define FWD_THREADS 512
-// V/K subtile: [128 rows][16 cols] bf16 + 16-B pad -> stride 4112:
-// 4112/4 = 1028 ≡ 4 (mod 32): staging ds_write_b128 lane groups spread
-// over distinct bank quads; 128-key tiles halve the barrier count per key
-// (the 64-key version parked 44% of wave time on rendezvous, PMC).
-#define FSUBT 4112
+// V/K subtile: [64 rows][16 cols] bf16 + 16-B pad -> stride 2080:
+// 2080/4 = 520 ≡ 8 (mod 32): the 8 staging ds_write_b128 of a lane group
+// hit 4 distinct bank quads (2-way); ≡ 8 (mod 64) keeps adjacent-subtile
+// tr blocks mostly disjoint.
+#define FSUBT 2080
 #define FWD_THR 11.54f  // defer-max threshold: 8 nats in log2 units
 #define LOG2E 1.44269504089f
 
@@ -74,32 +74,29 @@ __device__ inline int fst_addr(int tt, int row, int byte_in_row) {
 // iteration EARLY (under the previous tile's MFMA phase, hiding the HBM
 // latency); the writes land after the barrier that frees the buffer.
 struct KvRegs {
-  uint4 k[4], v[4];
+  uint4 k0, k1, v0, v1;
 };
 
-// thread t of 512 owns 16-B chunks {t, t+512, t+1024, t+1536} of the
-// [128 rows][16 chunks] flattened tile (chunk id -> row id>>4, col id&15)
 __device__ inline KvRegs load_kv(const bf16* __restrict__ ksrc,
                                  const bf16* __restrict__ vsrc, int64_t ld) {
+  const int t = threadIdx.x;
+  const int64_t off = (int64_t)(t >> 3) * ld + ((t & 7) * 2) * 8;
   KvRegs r;
-#pragma unroll
-  for (int rep = 0; rep < 4; rep++) {
-    const int id = threadIdx.x + rep * FWD_THREADS;
-    const int64_t off = (int64_t)(id >> 4) * ld + (id & 15) * 8;
-    r.k[rep] = reinterpret_cast<const uint4*>(ksrc + off)[0];
-    r.v[rep] = reinterpret_cast<const uint4*>(vsrc + off)[0];
-  }
+  r.k0 = reinterpret_cast<const uint4*>(ksrc + off)[0];
+  r.k1 = reinterpret_cast<const uint4*>(ksrc + off)[1];
+  r.v0 = reinterpret_cast<const uint4*>(vsrc + off)[0];
+  r.v1 = reinterpret_cast<const uint4*>(vsrc + off)[1];
   return r;
 }
 
 __device__ inline void write_kv(FwdTile* kd, FwdTile* vd, const KvRegs& r) {
-#pragma unroll
-  for (int rep = 0; rep < 4; rep++) {
-    const int id = threadIdx.x + rep * FWD_THREADS;
-    const int a0 = fst_addr((id & 15) >> 1, id >> 4, (id & 1) * 16);
-    *reinterpret_cast<uint4*>(kd->sub + a0) = r.k[rep];
-    *reinterpret_cast<uint4*>(vd->sub + a0) = r.v[rep];
-  }
+  const int t = threadIdx.x;
+  const int c0 = (t & 7) * 2;
+  const int a0 = fst_addr(c0 >> 1, t >> 3, (c0 & 1) * 16);
+  *reinterpret_cast<uint4*>(kd->sub + a0) = r.k0;
+  *reinterpret_cast<uint4*>(kd->sub + a0 + 16) = r.k1;
+  *reinterpret_cast<uint4*>(vd->sub + a0) = r.v0;
+  *reinterpret_cast<uint4*>(vd->sub + a0 + 16) = r.v1;
 }
 
 // K A-fragment: row = key (s*32 + l31), k-dim = d slice tt: contiguous b128
@@ -251,7 +248,7 @@ __global__ __launch_bounds__(FWD_THREADS, 1) void fa_fwd_kernel(
       const unsigned v_img_base = (unsigned)(uintptr_t)sm.v_img[cur].sub + tr_off;
 
 #pragma unroll
-      for (int s = 0; s < 4; s++) {  // four 32-key blocks per tile
+      for (int s = 0; s < 2; s++) {  // two 32-key blocks per tile
         const int kb0 = key0 + s * 32;
         if (causal && kb0 > my_q0 + FWD_QBLK - 1) break;
 

@@ -145,8 +145,9 @@ def main() -> None:
     dtype = torch.bfloat16
     if args.checkpoint_activations == "auto":
         # 288 GB HBM3E: 8B-class activations (~2 GB/layer at bs2 x 8k) fit
-        # without recompute; 70B needs checkpointing
-        ckpt = args.model == "llama3_70b"
+        # without recompute; 70B needs checkpointing, and the fp8-quantized
+        # allreduce path needs the headroom its wire buffers consume
+        ckpt = args.model == "llama3_70b" or args.quantize
     else:
         ckpt = args.checkpoint_activations == "on"
     model = Llama(cfg, dtype=dtype, checkpoint_activations=ckpt)

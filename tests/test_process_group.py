@@ -17,6 +17,7 @@ from torch.distributed.distributed_c10d import (
 )
 
 from torchft_amd.process_group import (
+    ProcessGroupWrapper,
     ErrorSwallowingProcessGroupWrapper,
     FakeProcessGroupWrapper,
     ProcessGroupDummy,
@@ -210,3 +211,46 @@ class TestAbortDiagnostics:
         monkeypatch.delenv("TORCHFT_ABORT_DUMP_DIR", raising=False)
         pg = ProcessGroupGloo(timeout=timedelta(seconds=5))
         assert pg._oplog.dump("x") is None
+
+
+class TestBackgroundReaper:
+    def test_configure_overlaps_old_abort(self):
+        """configure() must not block on the old communicator's teardown
+        (on MI355X the RCCL abort alone is ~507 ms) and must join the
+        previous teardown before starting the next."""
+        import time
+
+        aborted = []
+
+        class _SlowAbortPG:
+            def __init__(self, tag):
+                self.tag = tag
+
+            def abort(self):
+                time.sleep(0.3)
+                aborted.append(self.tag)
+
+        built = []
+
+        class _PG(ProcessGroupWrapper):
+            def _build(self, store, rank, world_size):
+                built.append(len(built))
+                return _SlowAbortPG(built[-1])
+
+        from torch.distributed import TCPStore
+
+        store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+        pg = _PG(timeout=timedelta(seconds=5))
+        pg.configure(f"127.0.0.1:{store.port}/ra", "r0", 0, 1)  # no old comm
+
+        t0 = time.perf_counter()
+        pg.configure(f"127.0.0.1:{store.port}/rb", "r0", 0, 1)
+        elapsed = time.perf_counter() - t0
+        assert elapsed < 0.25, f"configure blocked on the old abort ({elapsed:.2f}s)"
+        assert pg._reaper is not None
+
+        # third configure joins the in-flight teardown first
+        pg.configure(f"127.0.0.1:{store.port}/rc", "r0", 0, 1)
+        assert aborted == [0]  # first comm retired exactly once, in order
+        pg.shutdown()
+        assert aborted == [0, 1]

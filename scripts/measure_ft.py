@@ -34,6 +34,12 @@ def measure_reconfigure(n: int = 5) -> dict:
     pg = ProcessGroupRCCL(timeout=timedelta(seconds=60)) if use_cuda else ProcessGroupGloo()
     times = []
     for i in range(n + 1):
+        # membership changes are seconds apart in practice; give the
+        # background reaper time to retire the previous communicator, or
+        # this tight loop measures the old comm's ~500 ms abort (the join
+        # at the top of configure) instead of the rebuild
+        if pg._reaper is not None:
+            pg._reaper.join()
         t0 = time.perf_counter()
         pg.configure(f"127.0.0.1:{store.port}/reconf_{i}", "r0", 0, 1)
         # first collective proves the communicator is live

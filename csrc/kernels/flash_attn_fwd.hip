@@ -247,32 +247,17 @@ __global__ __launch_bounds__(FWD_THREADS, 1) void fa_fwd_kernel(
       const unsigned char* kimg = sm.k_img[cur].sub;
       const unsigned v_img_base = (unsigned)(uintptr_t)sm.v_img[cur].sub + tr_off;
 
-      // s-block software pipeline: block 1's QK^T chain (pure MFMA,
-      // independent of block 0's softmax state) issues BEFORE block 0's
-      // softmax, so the VALU-dense softmax overlaps matrix-pipe work
-      // instead of leaving it idle.
-      const bool vis1 = !(causal && key0 + 32 > my_q0 + FWD_QBLK - 1);
-      f32x16 s_pipe[2];
-      s_pipe[0] = f32x16{};
-#pragma unroll
-      for (int tt = 0; tt < 8; tt++) {
-        s_pipe[0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            k_afrag(kimg, tt, 0, half, l31), qfrag[tt], s_pipe[0], 0, 0, 0);
-      }
-      if (vis1) {
-        s_pipe[1] = f32x16{};
-#pragma unroll
-        for (int tt = 0; tt < 8; tt++) {
-          s_pipe[1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              k_afrag(kimg, tt, 1, half, l31), qfrag[tt], s_pipe[1], 0, 0, 0);
-        }
-      }
-
 #pragma unroll
       for (int s = 0; s < 2; s++) {  // two 32-key blocks per tile
         const int kb0 = key0 + s * 32;
-        if (s == 1 && !vis1) break;
-        const f32x16 s_acc = s_pipe[s];
+        if (causal && kb0 > my_q0 + FWD_QBLK - 1) break;
+
+        f32x16 s_acc = {};
+#pragma unroll
+        for (int tt = 0; tt < 8; tt++) {
+          s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              k_afrag(kimg, tt, s, half, l31), qfrag[tt], s_acc, 0, 0, 0);
+        }
 
         // raw scores + causal mask; per-lane block max over its q.
         // One multiply converts the max to the log2 domain (c2 > 0

@@ -141,6 +141,11 @@ def _diloco_replica(
                     criterion(model(x), y).backward()
                     inner_opt.step()
                     local_batches += 1
+                    # the DiLoCo post-hook may have live-healed PAST the
+                    # target step (slow start under suite load); fire here
+                    # too or the loop exits with the failure never injected
+                    if fail_armed and manager.current_step() >= fail_at_step:
+                        raise InjectedFailure(f"fail replica {replica_id}")
             return {
                 "original": {
                     f"{i}_{k}": v.detach().clone()

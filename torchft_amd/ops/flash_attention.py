@@ -8,15 +8,17 @@ the stock backward is the single largest kernel of the Llama-8B step
 Constraints of the custom backward: bf16, head_dim=128, seq % 128 == 0,
 dropout 0. Anything else falls back to stock SDPA.
 
-Status (round 2, measured on MI355X at the 8B bench shape, interleaved
-A/B): fwd+bwd 12.69 ms custom vs 13.13 ms stock — the hand-written
-backward BEATS aotriton's, so it is ON by default (opt out with
-TORCHFT_AMD_CUSTOM_FA=0). What got it there, in order: XOR swizzle of the
-LDS images (21.2 -> 17.6 ms), ds_read_b64_tr_b16 hardware-transpose reads
-replacing the b16 transpose-scatter images (-> 16.0 ms), and 8-wave
-workgroups sharing the staged tile stream for 2 waves/SIMD occupancy
-(-> 12.7 ms; the 4-wave version ran 1 wave/SIMD with every dependent
-stall exposed).
+Status (end of round 2, measured on MI355X at the 8B bench shape,
+interleaved A/B): fwd+bwd 11.80 ms custom vs 13.27 ms stock (-11%) — the
+hand-written backward beats aotriton's on both kernels, so it is ON by
+default (opt out with TORCHFT_AMD_CUSTOM_FA=0). Levers in order: XOR
+swizzle of the LDS images (21.2 -> 17.6 ms), ds_read_b64_tr_b16
+hardware-transpose reads replacing the b16 transpose-scatter images
+(-> 16.0), 8-wave workgroups for 2 waves/SIMD occupancy (-> 12.7), native
+[B,S,H,D] layout + fused delta (-> 12.5 and no aten copy chain), 64-key
+dq tiles (-> 11.8). The hand-written FORWARD (fa_fwd) is numerically
+pinned against aten but ~0.65 ms behind it, so it stays opt-in
+(TORCHFT_AMD_CUSTOM_FA_FWD=1; full-custom 12.4 ms still beats stock).
 """
 
 from __future__ import annotations

@@ -102,9 +102,14 @@ def _replica_main(
             criterion = nn.MSELoss()
             while manager.current_step() < total_steps:
                 injector.check(replica_id, manager.current_step(), pg)
-                torch.manual_seed(manager.current_step())  # same data all replicas
-                x = torch.randn(4, 8)
-                y = torch.randn(4, 4)
+                # per-call generator: identical data on every replica even
+                # when thread-ranks interleave (the GLOBAL RNG is shared —
+                # seed-then-draw races produced different batches, and a
+                # replica committing its final step alone after the peer
+                # exited then diverged without a later heal)
+                gen = torch.Generator().manual_seed(manager.current_step())
+                x = torch.randn(4, 8, generator=gen)
+                y = torch.randn(4, 4, generator=gen)
                 opt.zero_grad()
                 loss = criterion(ddp(x), y)
                 loss.backward()

@@ -211,7 +211,12 @@ class TestDiLoCo:
             lh.shutdown()
 
     def test_recovery_after_failure(self):
-        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=1000)
+        # min_replicas=2: the final window must be a JOINT commit — when
+        # the faster replica exits, its heartbeat stops and the laggard
+        # could otherwise commit the last outer step alone with its own
+        # (replica-specific) data, a divergence no later heal repairs.
+        # The survivor simply waits out the restart (~2 s) at its sync.
+        lh = LighthouseServer(bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=1000)
         try:
             with ThreadPoolExecutor(max_workers=2) as ex:
                 futs = [

@@ -82,3 +82,44 @@ def test_assignment_invariants(q, group_rank, init_sync):
     assert len(stores) == 1
     primary_id = stores.pop().removeprefix("s_")
     assert results[primary_id].max_replica_rank is not None
+
+
+class TestParticipationFunction:
+    """Invariants of the pure participation function the Manager uses
+    (torchft_amd/manager.py::_participation_from)."""
+
+    @given(
+        replica_rank=st.integers(0, 15),
+        world=st.integers(1, 16),
+        max_rank=st.one_of(st.none(), st.integers(0, 15)),
+        max_world=st.integers(1, 16),
+        defer=st.booleans(),
+        min_size=st.integers(1, 16),
+    )
+    def test_invariants(self, replica_rank, world, max_rank, max_world, defer,
+                        min_size):
+        from torchft_amd.manager import WorldSizeMode, _participation_from
+
+        for mode in (WorldSizeMode.DYNAMIC, WorldSizeMode.FIXED_WITH_SPARES):
+            rank, count = _participation_from(
+                replica_rank=replica_rank,
+                replica_world_size=world,
+                max_replica_rank=max_rank,
+                max_world_size=max_world,
+                defer_healing=defer,
+                world_size_mode=mode,
+                min_replica_size=min_size,
+            )
+            # a participating rank always fits inside the numeric world
+            if rank is not None:
+                assert 0 <= rank < max(count, rank + 1)
+            assert count >= 0
+            if mode == WorldSizeMode.FIXED_WITH_SPARES:
+                # the numeric world never exceeds the fixed size, and any
+                # overflow rank is benched, never renumbered
+                assert count <= min_size
+                if rank is not None:
+                    assert rank < min_size
+            else:
+                src = (max_rank, max_world) if defer else (replica_rank, world)
+                assert (rank, count) == src

@@ -227,16 +227,12 @@ def main() -> None:
         ddp_model = DistributedDataParallel(manager, model)
 
     # ---- optimizer ---------------------------------------------------------
-    if shards > 1:
-        base_opt = torch.optim.AdamW(
-            model.parameters(), lr=3e-4, betas=(0.9, 0.95), weight_decay=0.1,
-            foreach=True,
-        )
-    else:
-        from torchft_amd.ops import FusedAdamW
+    # FusedAdamW handles both plain and FSDP2 DTensor (sharded) params —
+    # DTensors update on their local shards
+    from torchft_amd.ops import FusedAdamW
 
-        base_opt = FusedAdamW(model.parameters(), lr=3e-4, betas=(0.9, 0.95),
-                              weight_decay=0.1)
+    base_opt = FusedAdamW(model.parameters(), lr=3e-4, betas=(0.9, 0.95),
+                          weight_decay=0.1)
 
     if manager is not None:
         from torchft_amd.optim import OptimizerWrapper

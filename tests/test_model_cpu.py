@@ -131,3 +131,41 @@ class TestLlama70BConfig:
             m = Llama(LLAMA3_8B, dtype=torch.bfloat16)
         n = m.num_params()
         assert 7.5e9 < n < 8.5e9, f"8B config has {n/1e9:.1f}B params"
+
+
+class TestFusedAdamWDTensor:
+    def test_sharded_params_match_plain(self):
+        """FusedAdamW on DTensor (FSDP2-sharded) params must produce the
+        same values as on plain tensors (enables the HSDP fused path)."""
+        import os
+
+        import torch.distributed as dist
+        from torch.distributed.device_mesh import init_device_mesh
+        from torch.distributed.tensor import DTensor, Shard, distribute_tensor
+
+        from torchft_amd.ops import FusedAdamW
+
+        store = dist.TCPStore("127.0.0.1", 0, is_master=True,
+                              wait_for_workers=False)
+        dist.init_process_group("gloo", store=store, rank=0, world_size=1)
+        try:
+            mesh = init_device_mesh("cpu", (1,))
+            torch.manual_seed(17)
+            base = torch.randn(64, 8)
+            p_plain = base.clone().requires_grad_(True)
+            p_dt = torch.nn.Parameter(
+                distribute_tensor(base.clone(), mesh, [Shard(0)])
+            )
+            opt_a = FusedAdamW([p_plain], lr=0.05)
+            opt_b = FusedAdamW([p_dt], lr=0.05)
+            for step in range(4):
+                g = torch.randn(64, 8)
+                p_plain.grad = g.clone()
+                p_dt.grad = distribute_tensor(g.clone(), mesh, [Shard(0)])
+                opt_a.step()
+                opt_b.step()
+            torch.testing.assert_close(
+                p_dt.to_local(), p_plain.detach(), rtol=1e-5, atol=1e-6
+            )
+        finally:
+            dist.destroy_process_group()

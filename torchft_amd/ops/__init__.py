@@ -198,11 +198,24 @@ def swiglu_glu(gu: torch.Tensor) -> torch.Tensor:
 # ----------------------------------------------------------------- Fused AdamW
 
 
+def _local_view(t: torch.Tensor) -> torch.Tensor:
+    """The local shard of a DTensor (FSDP2 sharded params/grads), else t.
+    The update runs on the shard in place — mathematically identical for
+    elementwise optimizers, and what lets the HSDP path use the fused
+    kernel."""
+    try:
+        from torch.distributed.tensor import DTensor
+    except ImportError:  # pragma: no cover
+        return t
+    return t.to_local() if isinstance(t, DTensor) else t
+
+
 class FusedAdamW(torch.optim.Optimizer):
     """Single-kernel multi-tensor AdamW for bf16 params (fp32 moments).
 
     One launch per step updates every parameter (csrc/kernels/fused_adamw.hip);
-    falls back to torch's foreach AdamW path on CPU.
+    falls back to torch's foreach AdamW path on CPU. DTensor params (the
+    FSDP2 sharded path) are updated on their local shards.
     """
 
     def __init__(
@@ -229,14 +242,16 @@ class FusedAdamW(torch.optim.Optimizer):
                 if p.grad is None:
                     continue
                 state = self.state[p]
+                p_loc = _local_view(p)
+                g_loc = _local_view(p.grad)
                 if len(state) == 0:
                     state["step"] = 0
-                    state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
-                    state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
+                    state["exp_avg"] = torch.zeros_like(p_loc, dtype=torch.float32)
+                    state["exp_avg_sq"] = torch.zeros_like(p_loc, dtype=torch.float32)
                 state["step"] += 1
                 bucket = by_step.setdefault(state["step"], [[], [], [], []])
-                bucket[0].append(p)
-                bucket[1].append(p.grad)
+                bucket[0].append(p_loc)
+                bucket[1].append(g_loc)
                 bucket[2].append(state["exp_avg"])
                 bucket[3].append(state["exp_avg_sq"])
 

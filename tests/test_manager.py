@@ -340,3 +340,65 @@ class TestQuorumArgsPropagation:
             assert client._quorum.call_args.kwargs["init_sync"] is True
         finally:
             m.shutdown(wait=False)
+
+
+class TestManagerCounters:
+    """Accessors and counter persistence (reference: manager_test.py's
+    state_dict/load_state_dict and participation accessor coverage)."""
+
+    def test_accessors_before_any_quorum(self):
+        client = MagicMock()
+        m = make_manager(client)
+        try:
+            assert m.participating_rank() is None
+            assert m.num_participants() == 0
+            assert not m.is_participating()
+            assert m.current_step() == 0
+            assert m.batches_committed() == 0
+        finally:
+            m.shutdown(wait=False)
+
+    def test_counters_roundtrip(self):
+        client = MagicMock()
+        m = make_manager(client)
+        try:
+            m.load_state_dict({"step": 7, "batches_committed": 21})
+            assert m.current_step() == 7
+            assert m.batches_committed() == 21
+            assert m.state_dict() == {"step": 7, "batches_committed": 21}
+        finally:
+            m.shutdown(wait=False)
+
+    def test_batches_committed_accumulates_participants(self):
+        # every committed step adds num_participants (the global batch is
+        # sized by live replicas, not a constant)
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum()
+        client.should_commit.return_value = True
+        m = make_manager(client)
+        try:
+            for expected in (2, 4):
+                m.start_quorum()
+                m.allreduce(torch.ones(4)).wait()
+                assert m.num_participants() == 2
+                assert m.should_commit()
+                assert m.batches_committed() == expected
+                client._quorum.return_value = mock_quorum(
+                    max_step=m.current_step()
+                )
+        finally:
+            m.shutdown(wait=False)
+
+    def test_failed_commit_does_not_advance_counters(self):
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum()
+        client.should_commit.return_value = False
+        m = make_manager(client)
+        try:
+            m.start_quorum()
+            m.allreduce(torch.ones(4)).wait()
+            assert not m.should_commit()
+            assert m.current_step() == 0
+            assert m.batches_committed() == 0
+        finally:
+            m.shutdown(wait=False)

@@ -299,3 +299,38 @@ class TestSplitContainers:
         assert len(frags) == 4
         total = sum(p.numel() for p in model.parameters())
         assert sum(p.numel() for f in frags for p in f.parameters()) == total
+
+
+class TestSyncSchedule:
+    """The staggered stage/commit schedule is a pure function of
+    (local_step, window, delay) — every replica computes the same phases."""
+
+    def test_no_delay_stage_and_commit_same_step(self):
+        from torchft_amd.local_sgd import _SyncSchedule
+
+        s = _SyncSchedule(window=4, delay=0)
+        stages = [i for i in range(1, 9) if s.stages_now(i)]
+        commits = [i for i in range(1, 9) if s.commits_now(i)]
+        assert stages == [4]
+        assert commits == [4]
+
+    def test_delay_staggers_stage_before_commit(self):
+        from torchft_amd.local_sgd import _SyncSchedule
+
+        s = _SyncSchedule(window=6, delay=2)
+        stages = [i for i in range(1, 13) if s.stages_now(i)]
+        commits = [i for i in range(1, 13) if s.commits_now(i)]
+        # stage fires `delay` steps before the commit so the allreduce
+        # overlaps `delay` inner steps
+        assert stages == [4]
+        assert commits == [6]
+        assert commits[0] - stages[0] == 2
+
+    def test_frozen_pure_value(self):
+        import dataclasses
+
+        from torchft_amd.local_sgd import _SyncSchedule
+
+        s = _SyncSchedule(window=4, delay=1)
+        with pytest.raises(dataclasses.FrozenInstanceError):
+            s.window = 5  # type: ignore[misc]

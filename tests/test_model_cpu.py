@@ -169,3 +169,44 @@ class TestFusedAdamWDTensor:
             )
         finally:
             dist.destroy_process_group()
+
+
+class TestPackedGLU:
+    def test_swiglu_glu_matches_unpacked(self):
+        from torchft_amd.ops import swiglu_glu, swiglu_ref
+
+        torch.manual_seed(0)
+        gu = torch.randn(3, 5, 8, dtype=torch.float32, requires_grad=True)
+        out = swiglu_glu(gu)
+        f = gu.shape[-1] // 2
+        ref = swiglu_ref(gu.detach()[..., :f], gu.detach()[..., f:])
+        torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-5)
+        # backward must route gradients to both halves
+        out.sum().backward()
+        assert gu.grad is not None
+        assert gu.grad[..., :f].abs().sum() > 0
+        assert gu.grad[..., f:].abs().sum() > 0
+
+
+class TestRopeTables:
+    def test_shapes_and_first_position(self):
+        from torchft_amd.ops import rope_tables
+
+        cos, sin = rope_tables(seq_len=16, head_dim=8, theta=500000.0)
+        assert cos.shape == (16, 4) and sin.shape == (16, 4)
+        # position 0: cos=1, sin=0 for every frequency
+        torch.testing.assert_close(cos[0], torch.ones(4))
+        torch.testing.assert_close(sin[0], torch.zeros(4))
+
+    def test_rope_preserves_norm(self):
+        from torchft_amd.ops import rope, rope_ref, rope_tables
+
+        torch.manual_seed(1)
+        cos, sin = rope_tables(seq_len=12, head_dim=16)
+        x = torch.randn(2, 12, 3, 16)
+        y = rope(x, cos, sin)
+        torch.testing.assert_close(y, rope_ref(x, cos, sin), rtol=1e-5, atol=1e-5)
+        # a rotation: per-pair L2 norms unchanged
+        torch.testing.assert_close(
+            y.pow(2).sum(-1), x.pow(2).sum(-1), rtol=1e-4, atol=1e-4
+        )

@@ -93,3 +93,49 @@ class TestLauncherComponent:
         assert env["REPLICA_GROUP_ID"] == "1"
         assert env["NUM_REPLICA_GROUPS"] == "4"
         assert env["TORCHFT_LIGHTHOUSE"] == "http://lh:123"
+
+
+class TestTelemetrySetup:
+    def test_file_handler_writes_json_lines(self, tmp_path):
+        import json
+        import logging
+
+        from torchft_amd.telemetry import STRUCTURED_LOGGERS, setup_telemetry
+
+        path = str(tmp_path / "telemetry.jsonl")
+        saved = {n: list(logging.getLogger(n).handlers) for n in STRUCTURED_LOGGERS}
+        try:
+            setup_telemetry(path)
+            log = logging.getLogger("torchft_quorums")
+            log.info("", extra={"replica_id": "r0", "quorum_id": 3, "step": 5})
+            for h in log.handlers:
+                h.flush()
+            lines = [json.loads(l) for l in open(path) if l.strip()]
+            assert len(lines) == 1
+            rec = lines[0]
+            assert rec["logger"] == "torchft_quorums"
+            assert rec["replica_id"] == "r0"
+            assert rec["quorum_id"] == 3 and rec["step"] == 5
+            assert "ts" in rec
+        finally:
+            for n in STRUCTURED_LOGGERS:
+                lg = logging.getLogger(n)
+                for h in list(lg.handlers):
+                    if h not in saved[n]:
+                        h.close()
+                        lg.removeHandler(h)
+                lg.propagate = True
+
+    def test_unserializable_extra_becomes_repr(self):
+        import json
+        import logging
+
+        from torchft_amd.telemetry import JSONLineFormatter
+
+        rec = logging.LogRecord(
+            "torchft_errors", logging.INFO, __file__, 1, "boom", (), None
+        )
+        rec.payload = object()  # not JSON-serializable
+        out = json.loads(JSONLineFormatter().format(rec))
+        assert out["msg"] == "boom"
+        assert out["payload"].startswith("<object object")

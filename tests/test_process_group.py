@@ -254,3 +254,57 @@ class TestBackgroundReaper:
         assert aborted == [0]  # first comm retired exactly once, in order
         pg.shutdown()
         assert aborted == [0, 1]
+
+
+class TestOpLogRing:
+    """The abort post-mortem ring buffer (flight-recorder analog)."""
+
+    def test_wraparound_keeps_latest(self):
+        from torchft_amd.process_group import _OpLog
+
+        log = _OpLog(capacity=4)
+        seqs = [log.record(f"op{i}", [torch.zeros(1)], quorum_id=1) for i in range(10)]
+        snap = log.snapshot()
+        assert len(snap) == 4
+        assert [r["seq"] for r in snap] == seqs[-4:]
+        assert [r["op"] for r in snap] == ["op6", "op7", "op8", "op9"]
+
+    def test_mark_updates_status_even_after_wrap(self):
+        from torchft_amd.process_group import _OpLog
+
+        log = _OpLog(capacity=4)
+        seqs = [log.record("allreduce", [torch.zeros(1)], 1) for _ in range(6)]
+        log.mark(seqs[-1], "completed")
+        log.mark(seqs[0], "completed")  # already evicted: silent no-op
+        snap = log.snapshot()
+        assert snap[-1]["status"] == "completed"
+        assert all(r["status"] == "issued" for r in snap[:-1])
+
+    def test_concurrent_record_is_sequential(self):
+        import threading
+
+        from torchft_amd.process_group import _OpLog
+
+        log = _OpLog(capacity=256)
+        out = []
+
+        def worker():
+            for _ in range(50):
+                out.append(log.record("op", [], None))
+
+        threads = [threading.Thread(target=worker) for _ in range(4)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        assert sorted(out) == list(range(1, 201))
+        snap = log.snapshot()
+        assert [r["seq"] for r in snap] == list(range(1, 201))
+
+    def test_dump_unwritable_dir_is_nonfatal(self, monkeypatch):
+        from torchft_amd.process_group import ABORT_DUMP_DIR_ENV, _OpLog
+
+        monkeypatch.setenv(ABORT_DUMP_DIR_ENV, "/proc/definitely/not/writable")
+        log = _OpLog()
+        log.record("allreduce", [torch.zeros(2)], 1)
+        assert log.dump("t") is None  # logged, not raised

@@ -76,3 +76,26 @@ def test_train_diloco_example(lighthouse):
          "--fragment-sync-delay", "0"],
         lighthouse.address(),
     )
+
+
+@pytest.mark.timeout(300)
+def test_launcher_two_replica_groups():
+    """End-to-end: the launcher spawns 2 replica-group torchruns + a
+    lighthouse; both groups join the quorum and train to completion."""
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "torchft_amd.launcher",
+            "--replicas", "2", "--min-replicas", "2",
+            "--base-port", "29770",
+            "--", "examples/train_ddp.py",
+            "--steps", "3", "--batch", "8", "--comm-stress-mb", "1",
+        ],
+        cwd=REPO,
+        capture_output=True,
+        text=True,
+        timeout=280,
+    )
+    assert proc.returncode == 0, (
+        f"launcher failed:\n{proc.stderr[-3000:]}\n{proc.stdout[-1000:]}"
+    )
+    assert "lighthouse at" in proc.stdout

@@ -80,6 +80,54 @@ class TestParameterServer:
         finally:
             ps.shutdown()
 
+    def test_unknown_path_404(self):
+        import urllib.error
+        import urllib.request
+
+        from torchft_amd.parameter_server import ParameterServer
+        from torchft_amd.process_group import ProcessGroup
+
+        class NopPS(ParameterServer):
+            def forward(self, session_id: str, pg: ProcessGroup) -> None:
+                pass
+
+        ps = NopPS()
+        try:
+            with pytest.raises(urllib.error.HTTPError):
+                urllib.request.urlopen(f"{ps.address()}/bogus", data=b"", timeout=10)
+        finally:
+            ps.shutdown()
+
+    def test_two_concurrent_sessions(self):
+        from concurrent.futures import ThreadPoolExecutor
+
+        from torchft_amd.parameter_server import ParameterServer
+        from torchft_amd.process_group import ProcessGroup
+
+        class EchoPS(ParameterServer):
+            def forward(self, session_id: str, pg: ProcessGroup) -> None:
+                # echo the client's tensor back doubled
+                t = torch.zeros(4)
+                pg.recv([t], 1, tag=0).wait()
+                pg.send([t * 2], 1, tag=1).wait()
+
+        ps = EchoPS()
+        try:
+            def client(val: float) -> torch.Tensor:
+                pg = EchoPS.connect(ps.address(), timeout=timedelta(seconds=20))
+                pg.send([torch.full((4,), val)], 0, tag=0).wait()
+                out = torch.zeros(4)
+                pg.recv([out], 0, tag=1).wait()
+                return out
+
+            with ThreadPoolExecutor(2) as ex:
+                f1 = ex.submit(client, 3.0)
+                f2 = ex.submit(client, 5.0)
+                torch.testing.assert_close(f1.result(timeout=60), torch.full((4,), 6.0))
+                torch.testing.assert_close(f2.result(timeout=60), torch.full((4,), 10.0))
+        finally:
+            ps.shutdown()
+
 
 class TestLauncherComponent:
     def test_build_replica_cmd(self):

@@ -47,6 +47,56 @@ class TestFutures:
         assert fired == []
 
 
+class TestTimerManagerLoad:
+    def test_callback_exception_does_not_kill_timer_thread(self):
+        import time as _time
+        from datetime import timedelta
+
+        from torchft_amd.futures import context_timeout
+
+        fired = []
+
+        def bad():
+            raise RuntimeError("callback bug")
+
+        # a raising callback must not wedge the shared timer loop
+        try:
+            with context_timeout(bad, timedelta(milliseconds=10)):
+                _time.sleep(0.1)
+        except RuntimeError:
+            pass
+        with context_timeout(lambda: fired.append(1), timedelta(milliseconds=20)):
+            _time.sleep(0.2)
+        assert fired == [1]
+
+    def test_many_concurrent_future_timeouts(self):
+        from datetime import timedelta
+
+        import torch
+
+        from torchft_amd.futures import future_timeout
+
+        completed = []
+        timed_out = []
+        futs = []
+        for i in range(40):
+            f: torch.futures.Future = torch.futures.Future()
+            tf = future_timeout(f, timedelta(milliseconds=150))
+            futs.append((i, f, tf))
+        # complete the even ones immediately; let the odd ones time out
+        for i, f, _ in futs:
+            if i % 2 == 0:
+                f.set_result(i)
+        for i, _, tf in futs:
+            try:
+                completed.append((i, tf.wait()))
+            except TimeoutError:
+                timed_out.append(i)
+        assert [i for i, v in completed] == [i for i in range(40) if i % 2 == 0]
+        assert all(v == i for i, v in completed)
+        assert timed_out == [i for i in range(40) if i % 2 == 1]
+
+
 class TestTelemetry:
     def test_json_formatter_includes_extras(self):
         fmt = JSONLineFormatter()

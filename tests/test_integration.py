@@ -73,6 +73,7 @@ def _replica_main(
     injector: EventInjector,
     total_steps: int,
     attempts: int = 3,
+    step_hook=None,  # called with the current step at the top of each loop
 ) -> Dict[str, torch.Tensor]:
     """Run one replica group (world_size=1) to total_steps, restarting on
     injected failures, and return the final model state dict."""
@@ -102,6 +103,8 @@ def _replica_main(
             criterion = nn.MSELoss()
             while manager.current_step() < total_steps:
                 injector.check(replica_id, manager.current_step(), pg)
+                if step_hook is not None:
+                    step_hook(manager.current_step())
                 # per-call generator: identical data on every replica even
                 # when thread-ranks interleave (the GLOBAL RNG is shared —
                 # seed-then-draw races produced different batches, and a
@@ -172,4 +175,63 @@ class TestFTIntegration:
         injector = EventInjector().fail_at(replica=0, step=2).fail_at(replica=2, step=3)
         dicts = _run_replicas(3, total_steps=7, injector=injector)
         assert injector.count == 2
+        assert_state_dicts_equal(dicts)
+
+
+class TestUpscale:
+    def test_late_joiner_heals_into_running_job(self):
+        """Reference analogue: local_sgd_integ_test's streaming upscale —
+        a third replica starts after the job is underway, live-heals to the
+        quorum's max step, and finishes bitwise-identical.
+
+        Deterministic rendezvous (no sleeps): the first two replicas block
+        at step 2 until the joiner's heartbeat shows up on the lighthouse
+        status page, so the join always overlaps the running job.
+        """
+        import urllib.request
+
+        lh = LighthouseServer(
+            bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=1000
+        )
+        injector = EventInjector()
+        total_steps = 8
+        reached_step2 = threading.Event()
+        joiner_seen = threading.Event()
+
+        def incumbent_hook(step: int) -> None:
+            if step >= 2:
+                reached_step2.set()
+                assert joiner_seen.wait(60), "joiner heartbeat never appeared"
+
+        def poll_for_joiner() -> None:
+            status_url = lh.address() + "/status"
+            while not joiner_seen.is_set():
+                try:
+                    with urllib.request.urlopen(status_url, timeout=5) as r:
+                        if "replica_2" in r.read().decode():
+                            joiner_seen.set()
+                            return
+                except OSError:
+                    pass
+                threading.Event().wait(0.05)
+
+        poller = threading.Thread(target=poll_for_joiner, daemon=True)
+        poller.start()
+        try:
+            with ThreadPoolExecutor(max_workers=3) as ex:
+                futs = [
+                    ex.submit(
+                        _replica_main, i, lh.address(), injector, total_steps,
+                        step_hook=incumbent_hook,
+                    )
+                    for i in range(2)
+                ]
+                assert reached_step2.wait(60)
+                futs.append(
+                    ex.submit(_replica_main, 2, lh.address(), injector, total_steps)
+                )
+                dicts = [f.result(timeout=120) for f in futs]
+        finally:
+            joiner_seen.set()  # unblock hooks on failure paths
+            lh.shutdown()
         assert_state_dicts_equal(dicts)

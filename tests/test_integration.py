@@ -62,9 +62,13 @@ class EventInjector:
             pg.report_future_error(RuntimeError("injected allreduce failure"))
 
 
-def _make_model() -> nn.Module:
-    torch.manual_seed(42)
-    return nn.Sequential(nn.Linear(8, 16), nn.ReLU(), nn.Linear(16, 4))
+def _make_model(seed: int = 42) -> nn.Module:
+    gen = torch.Generator().manual_seed(seed)
+    model = nn.Sequential(nn.Linear(8, 16), nn.ReLU(), nn.Linear(16, 4))
+    with torch.no_grad():
+        for p in model.parameters():
+            p.copy_(torch.randn(p.shape, generator=gen) * 0.1)
+    return model
 
 
 def _replica_main(
@@ -74,12 +78,13 @@ def _replica_main(
     total_steps: int,
     attempts: int = 3,
     step_hook=None,  # called with the current step at the top of each loop
+    model_seed: int = 42,
 ) -> Dict[str, torch.Tensor]:
     """Run one replica group (world_size=1) to total_steps, restarting on
     injected failures, and return the final model state dict."""
     for attempt in range(attempts):
         store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
-        model = _make_model()
+        model = _make_model(model_seed)
         pg = FakeProcessGroupWrapper(ProcessGroupGloo(timeout=timedelta(seconds=20)))
         manager = Manager(
             pg=pg,
@@ -233,5 +238,30 @@ class TestUpscale:
                 dicts = [f.result(timeout=120) for f in futs]
         finally:
             joiner_seen.set()  # unblock hooks on failure paths
+            lh.shutdown()
+        assert_state_dicts_equal(dicts)
+
+
+class TestInitSync:
+    def test_divergent_inits_converge_via_step0_sync(self):
+        """init_sync (default True) transfers step-0 state from the max-rank
+        replica: two replicas constructed with DIFFERENT weights must still
+        finish bitwise identical (reference: manager_integ_test's
+        skip-init-sync coverage, inverted)."""
+        lh = LighthouseServer(
+            bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=1000
+        )
+        injector = EventInjector()
+        try:
+            with ThreadPoolExecutor(max_workers=2) as ex:
+                futs = [
+                    ex.submit(
+                        _replica_main, i, lh.address(), injector, 5,
+                        model_seed=100 + i,  # deliberately different inits
+                    )
+                    for i in range(2)
+                ]
+                dicts = [f.result(timeout=120) for f in futs]
+        finally:
             lh.shutdown()
         assert_state_dicts_equal(dicts)

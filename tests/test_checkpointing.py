@@ -154,3 +154,58 @@ class TestHTTPTransport:
         finally:
             src.shutdown()
             dst.shutdown()
+
+
+class TestHTTPTransportErrorPaths:
+    def test_unreachable_source_raises_promptly(self):
+        import socket
+        import time
+        import urllib.error
+
+        # grab a port nothing listens on
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            dead_port = s.getsockname()[1]
+
+        dst = HTTPTransport(timeout=timedelta(seconds=5))
+        try:
+            t0 = time.monotonic()
+            with pytest.raises((urllib.error.URLError, OSError)):
+                dst.recv_checkpoint(
+                    src_rank=0,
+                    metadata=f"http://127.0.0.1:{dead_port}",
+                    step=1,
+                    timeout=timedelta(seconds=5),
+                )
+            assert time.monotonic() - t0 < 5.5  # bounded, not hung
+        finally:
+            dst.shutdown()
+
+    def test_fetch_retries_until_staged(self):
+        # the destination may issue its GET a beat before the source's
+        # quorum thread stages the snapshot; _open_with_retry absorbs the
+        # 400 window instead of failing the heal
+        import threading
+
+        src = HTTPTransport(timeout=timedelta(seconds=10))
+        dst = HTTPTransport(timeout=timedelta(seconds=10))
+        try:
+            sd = {"w": torch.arange(8.0)}
+
+            def stage_late():
+                threading.Event().wait(0.4)
+                src.send_checkpoint(
+                    [1], step=7, state_dict=sd, timeout=timedelta(seconds=10)
+                )
+
+            t = threading.Thread(target=stage_late)
+            t.start()
+            got = dst.recv_checkpoint(
+                src_rank=0, metadata=src.metadata(), step=7,
+                timeout=timedelta(seconds=10),
+            )
+            t.join()
+            torch.testing.assert_close(got["w"], sd["w"])
+        finally:
+            src.shutdown()
+            dst.shutdown()

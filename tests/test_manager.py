@@ -402,3 +402,65 @@ class TestManagerCounters:
             assert m.batches_committed() == 0
         finally:
             m.shutdown(wait=False)
+
+
+class TestOptimizerWrapper:
+    """zero_grad starts the quorum; step is gated by should_commit
+    (reference: torchft/optim.py semantics)."""
+
+    def _wrapped(self, client):
+        from torchft_amd.optim import OptimizerWrapper
+
+        m = make_manager(client)
+        lin = torch.nn.Linear(4, 2)
+        base = torch.optim.SGD(lin.parameters(), lr=0.1)
+        return m, lin, base, OptimizerWrapper(m, base)
+
+    def test_step_applies_only_on_commit(self):
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum()
+        client.should_commit.return_value = False
+        m, lin, base, opt = self._wrapped(client)
+        try:
+            before = {k: v.clone() for k, v in lin.state_dict().items()}
+            opt.zero_grad()
+            lin(torch.randn(3, 4)).sum().backward()
+            opt.step()  # vote fails -> parameters must not move
+            for k, v in lin.state_dict().items():
+                torch.testing.assert_close(v, before[k], rtol=0, atol=0)
+
+            client.should_commit.return_value = True
+            opt.zero_grad()
+            lin(torch.randn(3, 4)).sum().backward()
+            opt.step()  # commit -> parameters move
+            moved = any(
+                not torch.equal(v, before[k]) for k, v in lin.state_dict().items()
+            )
+            assert moved
+        finally:
+            m.shutdown(wait=False)
+
+    def test_zero_grad_starts_quorum(self):
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum()
+        m, lin, base, opt = self._wrapped(client)
+        try:
+            assert m.participating_rank() is None
+            opt.zero_grad()
+            m.wait_quorum()
+            assert client._quorum.called
+        finally:
+            m.shutdown(wait=False)
+
+    def test_state_dict_passthrough(self):
+        client = MagicMock()
+        m, lin, base, opt = self._wrapped(client)
+        try:
+            assert opt.state_dict() == base.state_dict()
+            opt.load_state_dict(base.state_dict())
+            assert opt.param_groups is base.param_groups
+            assert opt.state is base.state
+            with pytest.raises(AssertionError):
+                opt.step(closure=lambda: 0.0)
+        finally:
+            m.shutdown(wait=False)

@@ -99,3 +99,47 @@ class TestQuantizedAllreduceChoreography:
         # fixed-order reduce makes rank 0's and rank 1's outputs equal
         for a, b in zip(results[0], results[1]):
             assert torch.equal(a, b), "ranks diverged after quantized allreduce"
+
+    def test_world4_avg(self):
+        # AVG folds 1/world into the requantization scale; validate the
+        # same choreography at world 4 with residual-norm bounds
+        world = 4
+        torch.manual_seed(1)
+        inputs = {
+            r: [torch.randn(5000), torch.randn(100)] for r in range(world)
+        }
+        expected = [
+            sum(inputs[r][i] for r in range(world)) / world for i in range(2)
+        ]
+
+        store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+        addr = f"127.0.0.1:{store.port}/qar4"
+
+        def run(rank: int) -> List[torch.Tensor]:
+            tensors = [t.clone() for t in inputs[rank]]
+            pg = ProcessGroupGloo(timeout=timedelta(seconds=20))
+            try:
+                pg.configure(addr, f"r{rank}", rank, world)
+                _, _, _, slice_bytes = pack_geometry(tensors, world)
+                pack = quantize_pack_ref(tensors, world)
+                recv = torch.empty_like(pack)
+                pg.alltoall_base(recv, pack, [], [], AllToAllOptions()).wait()
+                my_slice = _reduce_slices_ref(recv, world, slice_bytes, avg=True)
+                pg.allgather_into_tensor_coalesced(
+                    [pack.view(-1)], [my_slice], AllgatherOptions()
+                ).wait()
+                return dequantize_pack_ref([t.numel() for t in tensors], pack, world)
+            finally:
+                pg.shutdown()
+
+        with ThreadPoolExecutor(max_workers=world) as ex:
+            results = list(ex.map(run, range(world)))
+
+        for rank_result in results:
+            for got, exp in zip(rank_result, expected):
+                # residual-norm bound: pointwise tolerances are the wrong
+                # shape for block-scaled fp8 (top-of-block ulp = amax/14)
+                assert (got - exp).norm() / exp.norm() < 0.08
+        for r in range(1, world):
+            for a, b in zip(results[0], results[r]):
+                assert torch.equal(a, b), "ranks diverged after quantized allreduce"

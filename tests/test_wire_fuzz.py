@@ -76,3 +76,44 @@ class TestWireFuzz:
         finally:
             mgr.shutdown()
             lh.shutdown()
+
+    def test_lighthouse_serves_while_connections_stall(self):
+        """Slowloris-style: stalled half-open connections must not block
+        real clients (per-connection threads, no single accept loop wedge)."""
+        lh = core.LighthouseServer(bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=10)
+        port = _port(lh.address())
+        conns = []
+        try:
+            for _ in range(5):
+                s = socket.create_connection(("127.0.0.1", port), timeout=5)
+                s.sendall(b"\x10")  # first byte of a length prefix, then stall
+                conns.append(s)
+            c = core.LighthouseClient(lh.address(), connect_timeout=timedelta(seconds=5))
+            c.heartbeat("live")
+            q = c.quorum(replica_id="live", timeout=timedelta(seconds=10))
+            assert len(q.participants) == 1
+        finally:
+            for s in conns:
+                s.close()
+            lh.shutdown()
+
+    def test_lighthouse_survives_random_payload_storm(self):
+        """Property-style fuzz: 30 random payloads (varied length, some with
+        plausible length prefixes) then a liveness check."""
+        lh = core.LighthouseServer(bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=10)
+        port = _port(lh.address())
+        rng = random.Random(1234)
+        try:
+            for i in range(30):
+                n = rng.randrange(0, 256)
+                body = bytes(rng.randrange(256) for _ in range(n))
+                if i % 3 == 0 and n >= 4:
+                    # plausible length prefix pointing past the real payload
+                    body = struct.pack("<I", rng.randrange(0, 1 << 20)) + body[4:]
+                _send_raw(port, body)
+            c = core.LighthouseClient(lh.address(), connect_timeout=timedelta(seconds=5))
+            c.heartbeat("storm_survivor")
+            q = c.quorum(replica_id="storm_survivor", timeout=timedelta(seconds=10))
+            assert len(q.participants) == 1
+        finally:
+            lh.shutdown()

@@ -140,3 +140,34 @@ class TestBabyGloo:
         finally:
             for pg in pgs:
                 pg.shutdown()
+
+
+class TestBabyAbort:
+    def test_abort_kills_child_and_reconfigure_recovers(self):
+        """abort() kills the worker subprocess (aborting any wedged
+        collective); a subsequent configure() respawns and works."""
+        store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+        pg = ProcessGroupBabyGloo(timeout=20.0)
+        try:
+            pg.configure(f"127.0.0.1:{store.port}/abort0", "r0", 0, 1)
+            t = torch.ones(4)
+            opts = AllreduceOptions()
+            opts.reduceOp = ReduceOp.SUM
+            pg.allreduce([t], opts).wait()
+
+            proc = pg._proc
+            assert proc is not None and proc.is_alive()
+            pg.abort()
+            proc.join(timeout=10)
+            assert not proc.is_alive()
+
+            # an op against the dead child must fail, not hang
+            with pytest.raises(Exception):
+                pg.allreduce([torch.ones(4)], opts).wait()
+
+            pg.configure(f"127.0.0.1:{store.port}/abort1", "r0", 0, 1)
+            t2 = torch.full((4,), 2.0)
+            pg.allreduce([t2], opts).wait()
+            torch.testing.assert_close(t2, torch.full((4,), 2.0))
+        finally:
+            pg.shutdown()

@@ -464,3 +464,39 @@ class TestOptimizerWrapper:
                 opt.step(closure=lambda: 0.0)
         finally:
             m.shutdown(wait=False)
+
+
+class TestPureDDP:
+    def test_grads_averaged_like_reducer_ddp(self):
+        """PureDDP is the per-parameter correctness oracle: its grads must
+        match the reducer-based DDP wrapper bitwise (both AVG over the
+        quorum's participant count)."""
+        from torchft_amd.ddp import (
+            DistributedDataParallel,
+            PureDistributedDataParallel,
+        )
+
+        def run(wrapper_cls):
+            client = MagicMock()
+            client._quorum.return_value = mock_quorum()  # 2 participants
+            client.should_commit.return_value = True
+            m = make_manager(client)
+            try:
+                torch.manual_seed(7)
+                model = torch.nn.Sequential(
+                    torch.nn.Linear(6, 8), torch.nn.ReLU(), torch.nn.Linear(8, 3)
+                )
+                wrapped = wrapper_cls(m, model)
+                m.start_quorum()
+                gen = torch.Generator().manual_seed(11)
+                x = torch.randn(4, 6, generator=gen)
+                wrapped(x).sum().backward()
+                assert m.should_commit()
+                return [p.grad.clone() for p in model.parameters()]
+            finally:
+                m.shutdown(wait=False)
+
+        pure = run(PureDistributedDataParallel)
+        reducer = run(DistributedDataParallel)
+        for gp, gr in zip(pure, reducer):
+            torch.testing.assert_close(gp, gr, rtol=0, atol=0)

@@ -77,7 +77,10 @@ class PureDistributedDataParallel(nn.Module):
 
         def reduce_grad(p: torch.Tensor) -> None:
             if p.grad is not None:
-                manager.allreduce(p.grad)
+                # wait() here: the managed work's continuation ledger (the
+                # AVG normalization) only replays once the work completes —
+                # dropping the handle would leave the gradient un-averaged
+                manager.allreduce(p.grad).wait()
 
         for p in module.parameters():
             p.register_post_accumulate_grad_hook(reduce_grad)

@@ -308,3 +308,55 @@ class TestOpLogRing:
         log = _OpLog()
         log.record("allreduce", [torch.zeros(2)], 1)
         assert log.dump("t") is None  # logged, not raised
+
+
+class TestDeadlineWork:
+    """_DeadlineWork guards the CPU-side wait: missing the deadline aborts
+    the process group (the RCCL watchdog-replacement policy), in user space
+    where the FT layer can observe it."""
+
+    class _SlowWork(torch.distributed._Work):
+        def __init__(self, block_s: float):
+            super().__init__()
+            self._block_s = block_s
+
+        def wait(self, timeout=None):
+            import time
+            time.sleep(self._block_s)
+            return True
+
+        def get_future(self):
+            fut = torch.futures.Future()
+            fut.set_result(None)
+            return fut
+
+    class _AbortRecorder:
+        def __init__(self):
+            self.aborted = 0
+
+        def abort(self):
+            self.aborted += 1
+
+    def test_expired_wait_aborts_pg(self):
+        from datetime import timedelta
+
+        from torchft_amd.process_group import _DeadlineWork
+
+        pg = self._AbortRecorder()
+        w = _DeadlineWork(pg, self._SlowWork(1.0), timedelta(milliseconds=100))
+        assert w.wait()  # the slow wait returns, but the deadline fired
+        import time
+        time.sleep(0.3)  # timer callback runs on the executor thread
+        assert pg.aborted >= 1
+
+    def test_fast_wait_does_not_abort(self):
+        from datetime import timedelta
+
+        from torchft_amd.process_group import _DeadlineWork
+
+        pg = self._AbortRecorder()
+        w = _DeadlineWork(pg, self._SlowWork(0.0), timedelta(seconds=5))
+        assert w.wait()
+        import time
+        time.sleep(0.2)
+        assert pg.aborted == 0

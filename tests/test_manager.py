@@ -500,3 +500,21 @@ class TestPureDDP:
         reducer = run(DistributedDataParallel)
         for gp, gr in zip(pure, reducer):
             torch.testing.assert_close(gp, gr, rtol=0, atol=0)
+
+
+class TestQuantizedFallback:
+    def test_should_quantize_falls_back_on_cpu(self):
+        # the fp8 path needs a HIP device; on CPU hosts the managed
+        # allreduce must silently use the plain wire path, not raise
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum()
+        client.should_commit.return_value = True
+        m = make_manager(client, should_quantize=True)
+        try:
+            m.start_quorum()
+            t = torch.ones(4)
+            assert m.allreduce(t).wait()
+            torch.testing.assert_close(t, torch.full((4,), 0.5))  # AVG by 2
+            assert m.should_commit()
+        finally:
+            m.shutdown(wait=False)

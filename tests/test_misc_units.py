@@ -169,3 +169,44 @@ class TestLighthouseCLI:
 
         with pytest.raises(SystemExit):
             lighthouse_main(["--min_replicas", "1", "--no-such-flag"])
+
+
+class TestPublicAPI:
+    def test_reference_surface_importable(self):
+        """The torchft public surface a migrating user needs (docs/MIGRATING.md)
+        must stay importable from the package root."""
+        import torchft_amd as ft
+
+        for name in [
+            "Manager", "WorldSizeMode", "OptimizerWrapper",
+            "DistributedDataParallel", "DistributedSampler",
+            "ProcessGroup", "ProcessGroupGloo", "ProcessGroupNCCL",
+            "ProcessGroupRCCL", "ProcessGroupDummy",
+            "ProcessGroupBabyGloo", "ProcessGroupBabyRCCL",
+            "ErrorSwallowingProcessGroupWrapper", "ManagedProcessGroup",
+            "DiLoCo", "LocalSGD", "split_into_fragments",
+        ]:
+            assert hasattr(ft, name), f"missing public export: {name}"
+            assert name in ft.__all__, f"{name} not in __all__"
+
+    def test_rccl_is_nccl_alias(self):
+        # on ROCm the "nccl" backend IS RCCL; both names must resolve to
+        # the same class so either spelling works in user code
+        import torchft_amd as ft
+
+        assert ft.ProcessGroupRCCL is ft.ProcessGroupNCCL
+
+    def test_submodule_surfaces(self):
+        from torchft_amd.checkpointing import HTTPTransport  # noqa: F401
+        from torchft_amd.checkpointing.pg_transport import PGTransport  # noqa: F401
+        from torchft_amd.collectives import (  # noqa: F401
+            allreduce_quantized,
+            reduce_scatter_quantized,
+        )
+        from torchft_amd.coordination import (  # noqa: F401
+            LighthouseClient,
+            LighthouseServer,
+        )
+        from torchft_amd.models import LLAMA3_8B, Llama  # noqa: F401
+        from torchft_amd.ops import FusedAdamW, RMSNorm  # noqa: F401
+        from torchft_amd.parallel.cp import ring_attention  # noqa: F401

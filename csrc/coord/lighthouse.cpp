@@ -327,7 +327,20 @@ void Lighthouse::handle_conn(int fd) {
                 state_.heartbeats[requester.replica_id] = Clock::now();
                 state_.participants[requester.replica_id] = {Clock::now(), requester};
               }
-              if (cv_.wait_until(lk, rpc_deadline) == std::cv_status::timeout) break;
+              if (Clock::now() >= rpc_deadline) break;
+              // Wait in slices shorter than the heartbeat timeout and
+              // refresh our own heartbeat each wake: a pending quorum
+              // request IS a liveness signal. Without this, a requester
+              // blocked here longer than heartbeat_timeout_ms (e.g. while
+              // straggler heartbeats block the split-brain guard) goes
+              // stale itself and can never be part of the quorum it is
+              // waiting for. (The participant entry is NOT refreshed —
+              // its `joined` time drives the join timeout.)
+              TimePoint wake = Clock::now() +
+                  Millis(std::max<int64_t>(opt_.heartbeat_timeout_ms / 4, 50));
+              if (wake > rpc_deadline) wake = rpc_deadline;
+              cv_.wait_until(lk, wake);
+              state_.heartbeats[requester.replica_id] = Clock::now();
             }
           }
           if (ok) {

@@ -397,3 +397,46 @@ class TestLighthouseFailover:
             mgr.shutdown()
             if lh2 is not None:
                 lh2.shutdown()
+
+
+class TestWaiterSelfHeartbeat:
+    def test_blocked_quorum_request_does_not_starve_itself(self):
+        """A replica blocked in the quorum RPC longer than
+        heartbeat_timeout_ms must not go stale itself: the pending request
+        counts as liveness (the waiter refreshes its own heartbeat each
+        wait slice). Regression: with 16 stale straggler heartbeats
+        blocking the split-brain guard, the waiter starved and timed out
+        even after the stragglers expired."""
+        import time
+        from concurrent.futures import ThreadPoolExecutor
+
+        lh = core.LighthouseServer(
+            bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=10
+        )
+        try:
+            def hammer(i):
+                c = core.LighthouseClient(
+                    lh.address(), connect_timeout=timedelta(seconds=10)
+                )
+                for _ in range(20):
+                    c.heartbeat(f"ghost_{i}")
+
+            with ThreadPoolExecutor(16) as ex:
+                list(ex.map(hammer, range(16)))
+
+            # request NOW, while the ghosts are still fresh: the guard
+            # blocks (1 participant vs 17 heartbeating) until they expire
+            # at heartbeat_timeout (5 s default) — then the waiter must
+            # still be healthy enough to form the quorum
+            c = core.LighthouseClient(
+                lh.address(), connect_timeout=timedelta(seconds=10)
+            )
+            c.heartbeat("survivor")
+            t0 = time.monotonic()
+            q = c.quorum(replica_id="survivor", timeout=timedelta(seconds=12))
+            elapsed = time.monotonic() - t0
+            assert len(q.participants) == 1
+            assert q.participants[0].replica_id == "survivor"
+            assert 4.0 < elapsed < 8.0  # formed right after ghost expiry
+        finally:
+            lh.shutdown()

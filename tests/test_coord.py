@@ -440,3 +440,86 @@ class TestWaiterSelfHeartbeat:
             assert 4.0 < elapsed < 8.0  # formed right after ghost expiry
         finally:
             lh.shutdown()
+
+
+class TestCommitBarrierStress:
+    def test_sixty_rounds_random_dissenter(self):
+        """should_commit is an all-ranks vote: any dissenting rank vetoes
+        the whole group, every rank sees the same verdict, and verdicts
+        from consecutive rounds never bleed into each other."""
+        import random
+
+        lh = core.LighthouseServer(
+            bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=100
+        )
+        mgr = core.ManagerServer(
+            replica_id="rep0", lighthouse_addr=lh.address(),
+            hostname="127.0.0.1", bind="127.0.0.1:0", store_addr="store0",
+            world_size=4, heartbeat_interval=TD(milliseconds=50),
+            connect_timeout=TD(seconds=5),
+        )
+        try:
+            rng = random.Random(0)
+            for step in range(1, 61):
+                dissenter = rng.randrange(5)  # 4 = unanimous round
+                results = [None] * 4
+
+                def vote(r):
+                    c = core.ManagerClient(
+                        mgr.address(), connect_timeout=TD(seconds=5)
+                    )
+                    results[r] = c.should_commit(
+                        r, step, r != dissenter, TD(seconds=10)
+                    )
+
+                ts = [threading.Thread(target=vote, args=(r,)) for r in range(4)]
+                [t.start() for t in ts]
+                [t.join(15) for t in ts]
+                expected = dissenter == 4
+                assert all(v == expected for v in results), (
+                    step, dissenter, results
+                )
+        finally:
+            mgr.shutdown()
+            lh.shutdown()
+
+
+class TestCoordResourceHygiene:
+    def test_server_lifecycle_no_fd_leak(self):
+        import os
+
+        def nfds():
+            return len(os.listdir(f"/proc/{os.getpid()}/fd"))
+
+        base = nfds()
+        for _ in range(10):
+            lh = core.LighthouseServer(
+                bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=10
+            )
+            c = core.LighthouseClient(lh.address(), connect_timeout=TD(seconds=5))
+            c.heartbeat("x")
+            q = c.quorum(replica_id="x", timeout=TD(seconds=5))
+            assert len(q.participants) == 1
+            lh.shutdown()
+        assert nfds() <= base + 5, "fd leak across server lifecycles"
+
+    def test_client_churn_no_fd_leak(self):
+        import os
+
+        def nfds():
+            return len(os.listdir(f"/proc/{os.getpid()}/fd"))
+
+        lh = core.LighthouseServer(
+            bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=10
+        )
+        try:
+            base = nfds()
+            for i in range(100):
+                c = core.LighthouseClient(
+                    lh.address(), connect_timeout=TD(seconds=5)
+                )
+                c.heartbeat(f"churn_{i % 4}")
+                del c
+            assert nfds() <= base + 5, "fd leak across client churn"
+        finally:
+            lh.shutdown()

@@ -518,3 +518,29 @@ class TestQuantizedFallback:
             assert m.should_commit()
         finally:
             m.shutdown(wait=False)
+
+
+class TestManagerLifecycleHygiene:
+    def test_repeated_construct_shutdown_no_fd_growth(self):
+        """Manager owns sockets, a serial-executor thread, and a checkpoint
+        transport; repeated lifecycle must not accumulate fds or threads."""
+        import gc
+        import os
+        import threading
+
+        def nfds():
+            return len(os.listdir(f"/proc/{os.getpid()}/fd"))
+
+        # warm-up cycle so lazily created module state is excluded
+        for i in range(2):
+            m = make_manager(MagicMock())
+            m.shutdown(wait=True)
+        gc.collect()
+        base_fds, base_thr = nfds(), threading.active_count()
+        for i in range(8):
+            m = make_manager(MagicMock())
+            m.shutdown(wait=True)
+            del m
+        gc.collect()
+        assert nfds() <= base_fds + 4, "fd growth across manager lifecycles"
+        assert threading.active_count() <= base_thr + 1

@@ -209,3 +209,43 @@ class TestHTTPTransportErrorPaths:
         finally:
             src.shutdown()
             dst.shutdown()
+
+
+class TestDTensorSerialization:
+    def test_dtensor_streaming_roundtrip(self):
+        """FSDP2 sharded state dicts hold DTensors; the streaming format
+        carries (local shard + pickled spec) and reconstructs the DTensor
+        (the HSDP heal path)."""
+        import io
+        import os
+
+        import torch.distributed as dist
+        from torch.distributed.device_mesh import init_device_mesh
+        from torch.distributed.tensor import DTensor, Shard, distribute_tensor
+
+        from torchft_amd.checkpointing._serialization import (
+            streaming_load,
+            streaming_save,
+        )
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29801")
+        created = not dist.is_initialized()
+        if created:
+            dist.init_process_group("gloo", rank=0, world_size=1)
+        try:
+            mesh = init_device_mesh("cpu", (1,))
+            full = torch.randn(8, 4)
+            dt = distribute_tensor(full, mesh, [Shard(0)])
+
+            buf = io.BytesIO()
+            streaming_save({"w": dt, "plain": torch.arange(4.0), "meta": 3}, buf)
+            buf.seek(0)
+            out = streaming_load(buf)
+            assert isinstance(out["w"], DTensor)
+            torch.testing.assert_close(out["w"].to_local(), dt.to_local())
+            torch.testing.assert_close(out["plain"], torch.arange(4.0))
+            assert out["meta"] == 3
+        finally:
+            if created:
+                dist.destroy_process_group()

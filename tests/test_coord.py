@@ -523,3 +523,40 @@ class TestCoordResourceHygiene:
             assert nfds() <= base + 5, "fd leak across client churn"
         finally:
             lh.shutdown()
+
+
+class TestQuorumUserData:
+    def test_member_data_round_trips(self):
+        """QuorumMember.data (the JSON user dict, reference
+        torchft.proto QuorumMember.data) must survive the quorum
+        round-trip so custom FT algorithms can piggyback metadata."""
+        import json
+
+        lh = core.LighthouseServer(
+            bind="127.0.0.1:0", min_replicas=2, join_timeout_ms=200
+        )
+        try:
+            payloads = {
+                "a": json.dumps({"zone": "az1", "slot": 3}),
+                "b": json.dumps({"zone": "az2"}),
+            }
+            results = {}
+
+            def join(rid):
+                c = core.LighthouseClient(
+                    lh.address(), connect_timeout=TD(seconds=5)
+                )
+                results[rid] = c.quorum(
+                    replica_id=rid, timeout=TD(seconds=10), data=payloads[rid]
+                )
+
+            ts = [threading.Thread(target=join, args=(r,)) for r in ("a", "b")]
+            [t.start() for t in ts]
+            [t.join(15) for t in ts]
+            assert set(results) == {"a", "b"}
+            for q in results.values():
+                by_id = {m.replica_id: m for m in q.participants}
+                assert json.loads(by_id["a"].data) == {"zone": "az1", "slot": 3}
+                assert json.loads(by_id["b"].data) == {"zone": "az2"}
+        finally:
+            lh.shutdown()

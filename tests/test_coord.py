@@ -560,3 +560,65 @@ class TestQuorumUserData:
                 assert json.loads(by_id["b"].data) == {"zone": "az2"}
         finally:
             lh.shutdown()
+
+
+class TestDashboardKill:
+    def test_kill_endpoint_terminates_manager(self):
+        """POST /replica/{id}/kill on the lighthouse forwards a kill RPC to
+        that replica's manager, which _exit(1)s (reference: the dashboard
+        kill button -> Manager.kill -> process::exit)."""
+        import subprocess
+        import sys
+        import urllib.request
+
+        lh = core.LighthouseServer(
+            bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=100
+        )
+        p = None
+        try:
+            child_code = f'''
+import time
+from datetime import timedelta as TD
+from torchft_amd import _ftcore as core
+mgr = core.ManagerServer(replica_id="victim", lighthouse_addr="{lh.address()}",
+    hostname="127.0.0.1", bind="127.0.0.1:0", store_addr="s", world_size=1,
+    heartbeat_interval=TD(milliseconds=50), connect_timeout=TD(seconds=5))
+c = core.ManagerClient(mgr.address(), connect_timeout=TD(seconds=5))
+c._quorum(group_rank=0, step=0, checkpoint_metadata="m", shrink_only=False,
+          timeout=TD(seconds=10))
+print("in quorum", flush=True)
+time.sleep(60)
+'''
+            p = subprocess.Popen(
+                [sys.executable, "-c", child_code],
+                stdout=subprocess.PIPE, text=True,
+            )
+            assert p.stdout.readline().strip() == "in quorum"
+            req = urllib.request.Request(
+                lh.address() + "/replica/victim/kill", data=b"", method="POST"
+            )
+            with urllib.request.urlopen(req, timeout=10) as r:
+                assert r.status == 200
+            assert p.wait(timeout=15) == 1  # manager _exit(1) on kill
+            p = None
+        finally:
+            if p is not None:
+                p.kill()
+                p.wait(timeout=10)
+            lh.shutdown()
+
+    def test_kill_unknown_replica_is_404(self):
+        import urllib.error
+        import urllib.request
+
+        lh = core.LighthouseServer(
+            bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=100
+        )
+        try:
+            req = urllib.request.Request(
+                lh.address() + "/replica/nobody/kill", data=b"", method="POST"
+            )
+            with pytest.raises(urllib.error.HTTPError):
+                urllib.request.urlopen(req, timeout=10)
+        finally:
+            lh.shutdown()

@@ -360,3 +360,46 @@ class TestDeadlineWork:
         import time
         time.sleep(0.2)
         assert pg.aborted == 0
+
+
+class TestCollectiveSurfaceGloo:
+    """Remaining collective-surface ops per backend (reference:
+    process_group_test.py's per-op table)."""
+
+    def test_broadcast_one(self):
+        def run(store_addr, rank, world):
+            pg = ProcessGroupGloo(timeout=timedelta(seconds=20))
+            pg.configure(store_addr, f"r{rank}", rank, world)
+            t = torch.full((4,), float(rank + 10))
+            pg.broadcast_one(t, root=1).wait()
+            return t
+
+        for t in _run_ranks(2, run):
+            torch.testing.assert_close(t, torch.full((4,), 11.0))
+
+    def test_allreduce_coalesced(self):
+        def run(store_addr, rank, world):
+            pg = ProcessGroupGloo(timeout=timedelta(seconds=20))
+            pg.configure(store_addr, f"r{rank}", rank, world)
+            ts = [torch.full((3,), float(rank + 1)), torch.full((5,), 2.0 * (rank + 1))]
+            from torch.distributed.distributed_c10d import AllreduceCoalescedOptions
+
+            opts = AllreduceCoalescedOptions()
+            opts.reduceOp = ReduceOp.SUM
+            pg.allreduce_coalesced(ts, opts).wait()
+            return ts
+
+        for ts in _run_ranks(2, run):
+            torch.testing.assert_close(ts[0], torch.full((3,), 3.0))
+            torch.testing.assert_close(ts[1], torch.full((5,), 6.0))
+
+    def test_barrier(self):
+        from torch.distributed.distributed_c10d import BarrierOptions
+
+        def run(store_addr, rank, world):
+            pg = ProcessGroupGloo(timeout=timedelta(seconds=20))
+            pg.configure(store_addr, f"r{rank}", rank, world)
+            pg.barrier(BarrierOptions()).wait()
+            return True
+
+        assert all(_run_ranks(2, run))

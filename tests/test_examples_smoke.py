@@ -99,3 +99,51 @@ def test_launcher_two_replica_groups():
         f"launcher failed:\n{proc.stderr[-3000:]}\n{proc.stdout[-1000:]}"
     )
     assert "lighthouse at" in proc.stdout
+
+
+@pytest.mark.timeout(120)
+def test_punisher_scrapes_and_kills():
+    """The chaos punisher discovers replicas by scraping the dashboard's
+    kill buttons and fires the kill endpoint — run it against a live
+    victim manager in a subprocess."""
+    import importlib.util
+
+    from torchft_amd import _ftcore as core
+
+    spec = importlib.util.spec_from_file_location(
+        "punisher", os.path.join(REPO, "examples", "chaos", "punisher.py")
+    )
+    punisher = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(punisher)
+
+    lh = core.LighthouseServer(bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=100)
+    p = None
+    try:
+        assert punisher.list_replicas(lh.address()) == []  # empty dashboard
+
+        child_code = f'''
+import time
+from datetime import timedelta as TD
+from torchft_amd import _ftcore as core
+mgr = core.ManagerServer(replica_id="victim", lighthouse_addr="{lh.address()}",
+    hostname="127.0.0.1", bind="127.0.0.1:0", store_addr="s", world_size=1,
+    heartbeat_interval=TD(milliseconds=50), connect_timeout=TD(seconds=5))
+c = core.ManagerClient(mgr.address(), connect_timeout=TD(seconds=5))
+c._quorum(group_rank=0, step=0, checkpoint_metadata="m", shrink_only=False,
+          timeout=TD(seconds=10))
+print("in quorum", flush=True)
+time.sleep(60)
+'''
+        p = subprocess.Popen(
+            [sys.executable, "-c", child_code], stdout=subprocess.PIPE, text=True
+        )
+        assert p.stdout.readline().strip() == "in quorum"
+        assert punisher.list_replicas(lh.address()) == ["victim"]
+        punisher.kill_one(lh.address())
+        assert p.wait(timeout=15) == 1
+        p = None
+    finally:
+        if p is not None:
+            p.kill()
+            p.wait(timeout=10)
+        lh.shutdown()

@@ -265,3 +265,56 @@ class TestInitSync:
         finally:
             lh.shutdown()
         assert_state_dicts_equal(dicts)
+
+
+class TestGroupWorld2:
+    def test_two_rank_group_quorum_and_commit(self):
+        """One replica group with world_size=2 (the HSDP shape): group rank
+        0 hosts the ManagerServer, rank 1 discovers it via the group store;
+        one quorum request covers both ranks and should_commit barriers
+        across them."""
+        from torchft_amd.manager import WorldSizeMode
+
+        lh = LighthouseServer(
+            bind="127.0.0.1:0", min_replicas=1, join_timeout_ms=200
+        )
+        store = TCPStore("127.0.0.1", 0, is_master=True, wait_for_workers=False)
+        try:
+            def run(rank: int):
+                state = {"w": torch.zeros(4)}
+                pg = ProcessGroupGloo(timeout=timedelta(seconds=20))
+                m = Manager(
+                    pg=pg,
+                    load_state_dict=state.update,
+                    state_dict=lambda: state,
+                    min_replica_size=1,
+                    rank=rank,
+                    world_size=2,
+                    store_addr="127.0.0.1",
+                    store_port=store.port,
+                    lighthouse_addr=lh.address(),
+                    replica_id="grp0",
+                    hostname="127.0.0.1",
+                    timeout=timedelta(seconds=20),
+                    quorum_timeout=timedelta(seconds=20),
+                    connect_timeout=timedelta(seconds=10),
+                )
+                try:
+                    for step in range(3):
+                        m.start_quorum()
+                        t = torch.full((4,), float(rank + 1))
+                        assert m.allreduce(t).wait()
+                        # single replica: cross-replica allreduce is world-1
+                        torch.testing.assert_close(
+                            t, torch.full((4,), float(rank + 1))
+                        )
+                        assert m.should_commit()
+                    assert m.current_step() == 3
+                    return True
+                finally:
+                    m.shutdown(wait=False)
+
+            with ThreadPoolExecutor(max_workers=2) as ex:
+                assert all(ex.map(run, range(2)))
+        finally:
+            lh.shutdown()

@@ -334,3 +334,38 @@ class TestSyncSchedule:
         s = _SyncSchedule(window=4, delay=1)
         with pytest.raises(dataclasses.FrozenInstanceError):
             s.window = 5  # type: ignore[misc]
+
+
+class TestSplitFragmentsProperties:
+    def test_partition_invariants(self):
+        """split_into_fragments partitions the model: every parameter in
+        exactly one fragment, fragment count as requested, no empties."""
+        from hypothesis import given, settings
+        from hypothesis import strategies as st
+
+        from torchft_amd.local_sgd import split_into_fragments
+
+        @settings(max_examples=30, deadline=None)
+        @given(
+            n_layers=st.integers(2, 12),
+            n_fragments=st.integers(1, 6),
+            width=st.integers(1, 8),
+        )
+        def run(n_layers, n_fragments, width):
+            if n_fragments > n_layers:
+                return
+            model = nn.Sequential(
+                *[nn.Linear(width, width) for _ in range(n_layers)]
+            )
+            frags = split_into_fragments(model, n_fragments)
+            assert len(frags) == n_fragments
+            seen = set()
+            for f in frags:
+                params = list(f.parameters())
+                assert params, "empty fragment"
+                for p in params:
+                    assert id(p) not in seen, "parameter in two fragments"
+                    seen.add(id(p))
+            assert len(seen) == len(list(model.parameters()))
+
+        run()

@@ -544,3 +544,25 @@ class TestManagerLifecycleHygiene:
         gc.collect()
         assert nfds() <= base_fds + 4, "fd growth across manager lifecycles"
         assert threading.active_count() <= base_thr + 1
+
+
+class TestGlobalRankDerivation:
+    def test_trailing_digits(self):
+        from torchft_amd.manager import extract_trailing_digits
+
+        assert extract_trailing_digits("replica_57") == 57
+        assert extract_trailing_digits("bench0") == 0
+        assert extract_trailing_digits("train_ddp_12") == 12
+        assert extract_trailing_digits("no_digits_") == 0
+        assert extract_trailing_digits("") == 0
+        assert extract_trailing_digits("123") == 123
+
+    def test_global_rank_formula(self):
+        # global_rank = replica_index * group_world_size + group_rank
+        # (reference: manager.py's trailing-digit derivation)
+        client = MagicMock()
+        m = make_manager(client, replica_id="replica_3")
+        try:
+            assert m._global_rank == 3 * 1 + 0
+        finally:
+            m.shutdown(wait=False)

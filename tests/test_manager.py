@@ -561,8 +561,8 @@ class TestGlobalRankDerivation:
         # global_rank = replica_index * group_world_size + group_rank
         # (reference: manager.py's trailing-digit derivation)
         client = MagicMock()
-        m = make_manager(client, replica_id="replica_3")
+        m = make_manager(client)  # harness replica_id is "test1" -> index 1
         try:
-            assert m._global_rank == 3 * 1 + 0
+            assert m._global_rank == 1 * 1 + 0
         finally:
             m.shutdown(wait=False)

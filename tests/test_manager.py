@@ -566,3 +566,39 @@ class TestGlobalRankDerivation:
             assert m._global_rank == 1 * 1 + 0
         finally:
             m.shutdown(wait=False)
+
+
+class TestManagedProcessGroup:
+    def test_routes_allreduce_and_size(self):
+        """ManagedProcessGroup adapts a Manager to the PG interface so
+        stock DDP/FSDP hooks get the FT allreduce; size() is the live
+        participant count (reference: ManagedProcessGroup semantics)."""
+        from torch.distributed import ReduceOp as RO
+        from torch.distributed.distributed_c10d import AllreduceOptions
+
+        from torchft_amd.process_group import ManagedProcessGroup
+
+        client = MagicMock()
+        client._quorum.return_value = mock_quorum()  # 2 participants
+        client.should_commit.return_value = True
+        m = make_manager(client)
+        try:
+            mpg = ManagedProcessGroup(m)
+            m.start_quorum()
+            assert mpg.size() == 2
+
+            t = torch.ones(4)
+            opts = AllreduceOptions()
+            opts.reduceOp = RO.AVG
+            assert mpg.allreduce([t], opts).wait()
+            torch.testing.assert_close(t, torch.full((4,), 0.5))  # AVG by 2
+
+            t2 = torch.ones(4)
+            assert mpg.allreduce([t2], RO.SUM).wait()  # bare-op form
+            torch.testing.assert_close(t2, torch.ones(4))
+
+            with pytest.raises(AssertionError):
+                mpg.allreduce([t, t2], opts)  # exactly one tensor
+            assert m.should_commit()
+        finally:
+            m.shutdown(wait=False)
